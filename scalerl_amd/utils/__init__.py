@@ -1,0 +1,11 @@
+from .logging import get_logger
+from .schedulers import LinearDecayScheduler, MultiStepScheduler, PiecewiseScheduler
+from .timings import Timings, Timer
+from .model_utils import hard_target_update, soft_target_update
+from .checkpoint import load_checkpoint, save_checkpoint
+
+__all__ = [
+    "get_logger", "LinearDecayScheduler", "MultiStepScheduler",
+    "PiecewiseScheduler", "Timings", "Timer", "hard_target_update",
+    "soft_target_update", "save_checkpoint", "load_checkpoint",
+]
