@@ -1,0 +1,180 @@
+"""Done-masked multi-layer LSTM core.
+
+Reference semantics: AtariNet's per-step unroll with done-masking
+(algorithms/utils/atari_model.py:109-120) — `state *= notdone` each step,
+then one LSTM step.  The reference calls nn.LSTM per step from Python; here:
+
+- input-side GEMM for ALL T steps is hoisted into one rocBLAS GEMM
+  ([T*B, in] @ W_ih^T);
+- the sequential loop does one rocBLAS GEMM (h @ W_hh^T) + ONE fused HIP
+  pointwise kernel per step (csrc/lstm.hip), forward and backward;
+- CPU path uses the same structure with torch ops (the oracle).
+
+Weights are plain nn.LSTM-layout parameters (weight_ih_l{k} etc.) so
+checkpoints interop with torch LSTMs.
+"""
+
+from __future__ import annotations
+
+import ctypes
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from . import _backend
+
+_c = ctypes.c_void_p
+
+
+def _pointwise_fwd(gates: torch.Tensor, c_prev: torch.Tensor):
+    """Returns (h, c); on GPU `gates` is overwritten with activated gates."""
+    B, H4 = gates.shape
+    H = H4 // 4
+    if gates.is_cuda:
+        h = torch.empty(B, H, device=gates.device, dtype=torch.float32)
+        c = torch.empty_like(h)
+        ret = _backend.lib().lstm_pointwise_fwd(
+            _c(gates.data_ptr()), _c(c_prev.data_ptr()), _c(h.data_ptr()),
+            _c(c.data_ptr()), B, H, _backend.current_stream())
+        _backend.check(ret, "lstm_pointwise_fwd")
+        return h, c
+    i, f, g, o = gates.chunk(4, dim=1)
+    i, f, g, o = i.sigmoid(), f.sigmoid(), g.tanh(), o.sigmoid()
+    c = f * c_prev + i * g
+    h = o * torch.tanh(c)
+    gates.copy_(torch.cat([i, f, g, o], dim=1))  # match GPU in-place contract
+    return h, c
+
+
+def _pointwise_bwd(gates_act, c_prev, c_out, dh, dc_in):
+    B, H4 = gates_act.shape
+    H = H4 // 4
+    if gates_act.is_cuda:
+        dgates = torch.empty_like(gates_act)
+        dc_prev = torch.empty(B, H, device=gates_act.device, dtype=torch.float32)
+        ret = _backend.lib().lstm_pointwise_bwd(
+            _c(gates_act.data_ptr()), _c(c_prev.data_ptr()), _c(c_out.data_ptr()),
+            _c(dh.data_ptr()), _c(dc_in.data_ptr()) if dc_in is not None else None,
+            _c(dgates.data_ptr()), _c(dc_prev.data_ptr()), B, H,
+            _backend.current_stream())
+        _backend.check(ret, "lstm_pointwise_bwd")
+        return dgates, dc_prev
+    i, f, g, o = gates_act.chunk(4, dim=1)
+    tc = torch.tanh(c_out)
+    dc = dh * o * (1 - tc * tc)
+    if dc_in is not None:
+        dc = dc + dc_in
+    di = dc * g * i * (1 - i)
+    df = dc * c_prev * f * (1 - f)
+    dg = dc * i * (1 - g * g)
+    do = dh * tc * o * (1 - o)
+    return torch.cat([di, df, dg, do], dim=1), dc * f
+
+
+class _MaskedLSTMFn(torch.autograd.Function):
+    """One LSTM layer unrolled over T with per-step done masking.
+
+    x: [T,B,I]; notdone: [T,B,1]; h0,c0: [B,H]; weights nn.LSTM layout.
+    Strategy: xW^T for all T hoisted to one GEMM; per step one GEMM + one
+    pointwise kernel.  Saves activated gates + cell states for backward.
+    """
+
+    @staticmethod
+    def forward(ctx, x, notdone, h0, c0, w_ih, w_hh, b_ih, b_hh):
+        T, B, I = x.shape
+        H = w_hh.shape[1]
+        xg = torch.addmm(b_ih + b_hh, x.reshape(T * B, I), w_ih.t()).view(T, B, 4 * H)
+        h, c = h0.contiguous(), c0.contiguous()
+        hs = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
+        cs_in = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
+        cs_out = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
+        hs_in = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
+        gates_all = torch.empty(T, B, 4 * H, device=x.device, dtype=x.dtype)
+        for t in range(T):
+            nd = notdone[t]
+            h = h * nd
+            c = c * nd
+            hs_in[t] = h
+            cs_in[t] = c
+            gates = torch.addmm(xg[t], h, w_hh.t())
+            gates_all[t] = gates
+            h, c = _pointwise_fwd(gates_all[t], cs_in[t])
+            hs[t] = h
+            cs_out[t] = c
+        ctx.save_for_backward(x, notdone, hs_in, cs_in, cs_out, gates_all,
+                              w_ih, w_hh)
+        ctx.H = H
+        return hs, h, c
+
+    @staticmethod
+    def backward(ctx, d_hs, d_hT, d_cT):
+        (x, notdone, hs_in, cs_in, cs_out, gates_all, w_ih,
+         w_hh) = ctx.saved_tensors
+        T, B, I = x.shape
+        H = ctx.H
+        dh_carry = d_hT.contiguous()
+        dc_carry = d_cT.contiguous()
+        dgates_all = torch.empty_like(gates_all)
+        for t in range(T - 1, -1, -1):
+            dh = d_hs[t] + dh_carry
+            dgates, dc_prev = _pointwise_bwd(gates_all[t], cs_in[t], cs_out[t],
+                                             dh.contiguous(), dc_carry)
+            dgates_all[t] = dgates
+            nd = notdone[t]
+            dh_carry = (dgates @ w_hh) * nd
+            dc_carry = dc_prev * nd
+        dg2 = dgates_all.reshape(T * B, 4 * H)
+        dx = (dg2 @ w_ih).view(T, B, I)
+        dw_ih = dg2.t() @ x.reshape(T * B, I)
+        dw_hh = dg2.t() @ hs_in.reshape(T * B, H)
+        db = dg2.sum(0)
+        return dx, None, dh_carry, dc_carry, dw_ih, dw_hh, db, db
+
+
+class MaskedLSTM(nn.Module):
+    """Multi-layer done-masked LSTM with nn.LSTM-compatible parameters."""
+
+    def __init__(self, input_size: int, hidden_size: int, num_layers: int = 1):
+        super().__init__()
+        self.input_size, self.hidden_size = input_size, hidden_size
+        self.num_layers = num_layers
+        for k in range(num_layers):
+            in_sz = input_size if k == 0 else hidden_size
+            w_ih = nn.Parameter(torch.empty(4 * hidden_size, in_sz))
+            w_hh = nn.Parameter(torch.empty(4 * hidden_size, hidden_size))
+            b_ih = nn.Parameter(torch.zeros(4 * hidden_size))
+            b_hh = nn.Parameter(torch.zeros(4 * hidden_size))
+            for w in (w_ih, w_hh):
+                nn.init.uniform_(w, -hidden_size ** -0.5, hidden_size ** -0.5)
+            setattr(self, f"weight_ih_l{k}", w_ih)
+            setattr(self, f"weight_hh_l{k}", w_hh)
+            setattr(self, f"bias_ih_l{k}", b_ih)
+            setattr(self, f"bias_hh_l{k}", b_hh)
+
+    def initial_state(self, batch_size: int, device=None):
+        z = torch.zeros(self.num_layers, batch_size, self.hidden_size,
+                        device=device)
+        return (z, z.clone())
+
+    def forward(self, x: torch.Tensor, notdone: torch.Tensor,
+                state: Tuple[torch.Tensor, torch.Tensor]):
+        """x [T,B,I], notdone [T,B] → out [T,B,H], (hN, cN) [L,B,H]."""
+        h0, c0 = state
+        # The recurrent core runs fp32 (it is tiny and latency-bound; the
+        # fused pointwise kernels are fp32) even when the encoder is bf16.
+        x = x.float()
+        nd = notdone.unsqueeze(-1).float()
+        hs_out: List[torch.Tensor] = []
+        cs_out: List[torch.Tensor] = []
+        out = x
+        for k in range(self.num_layers):
+            out, hN, cN = _MaskedLSTMFn.apply(
+                out.contiguous(), nd, h0[k], c0[k],
+                getattr(self, f"weight_ih_l{k}"),
+                getattr(self, f"weight_hh_l{k}"),
+                getattr(self, f"bias_ih_l{k}"),
+                getattr(self, f"bias_hh_l{k}"))
+            hs_out.append(hN)
+            cs_out.append(cN)
+        return out, (torch.stack(hs_out), torch.stack(cs_out))
