@@ -1,0 +1,147 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: IMPALA on synthetic Atari (BASELINE.json config 3).
+
+Measures env-steps/sec (whole job, all ranks): one timed "step" is one
+learner iteration consuming rollout_length × batch_size env steps pulled
+from live actor processes (the reference's SPS definition,
+impala_atari.py:391).  The full pipeline runs during timing: CPU actor
+processes stepping vectorized synthetic envs, policy inference, shared-
+memory rollout transport, H2D side-stream copies, learner fwd/bwd with the
+fused HIP V-trace/loss kernels, RCCL flat-grad all-reduce, fused RMSProp,
+and weight publication back to the actors.
+
+Single node, one rank per GPU (launched by torch.distributed.run for
+N > 1); weak scaling — per-rank actor count and batch are fixed.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--rollout-length", type=int, default=80)
+    p.add_argument("--batch-size", type=int, default=32)
+    p.add_argument("--envs-per-actor", type=int, default=16)
+    p.add_argument("--num-actors", type=int, default=0,
+                   help="actor procs per rank (0 = auto from cpu count)")
+    p.add_argument("--use-lstm", type=int, default=1)
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp32"])
+    p.add_argument("--device", type=str, default="auto")
+    p.add_argument("--inference", type=str, default="auto",
+                   choices=["auto", "cpu", "gpu"])
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_cuda = torch.cuda.is_available() and args.device in ("auto", "cuda")
+    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+
+    if args.num_actors <= 0:
+        ncpu = os.cpu_count() or 8
+        # leave ~2 cores per rank for the learner process + OS
+        args.num_actors = max(2, (ncpu - 2 * world) // max(1, world))
+
+    from scalerl_amd.config import ImpalaArguments
+    from scalerl_amd.runtime.impala import ImpalaTrainer
+
+    inference = args.inference
+    if inference == "auto":
+        inference = "gpu" if use_cuda else "cpu"
+
+    cfg = ImpalaArguments(
+        rollout_length=args.rollout_length, batch_size=args.batch_size,
+        envs_per_actor=args.envs_per_actor, num_actors=args.num_actors,
+        use_lstm=bool(args.use_lstm), device=device, dtype=args.dtype,
+        inference=inference, seed=1234 + rank,
+        total_steps=1 << 60, disable_checkpoint=True,
+        output_dir="/tmp/scalerl_bench")
+
+    # Build trainer (forks actor processes) BEFORE any CUDA/HIP init.
+    trainer = ImpalaTrainer(cfg, device=device)
+    trainer.rank = rank
+    trainer.start_actors()
+
+    # Now the device side (CUDA ctx + RCCL process group).
+    if world > 1:
+        from scalerl_amd.parallel.dist import init_distributed
+        init_distributed("nccl" if use_cuda else "gloo")
+    trainer.setup_learner()
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.train_iteration()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.train_iteration()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (the slowest rank defines whole-job time)
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    steps_per_iter = args.rollout_length * args.batch_size
+    total_env_steps = args.steps * steps_per_iter * world
+    value = total_env_steps / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "env_steps_per_sec",
+            "value": round(value, 1),
+            "unit": "env-steps/s",
+            "n_gpus": world if use_cuda else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if use_cuda else "fp32",
+            "data": "synthetic 84x84x4 frames, random-init weights",
+            "config": {
+                "model": "IMPALA AtariNet (3conv+FC512+2xLSTM519)",
+                "algo": "impala-vtrace",
+                "rollout_length": args.rollout_length,
+                "global_batch": args.batch_size * world,
+                "seq_len": args.rollout_length,
+                "envs_per_actor": args.envs_per_actor,
+                "actors_per_rank": args.num_actors,
+                "inference": inference,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    trainer.shutdown()
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -50,7 +50,9 @@ class SyntheticAtariEnv(Env):
     def step(self, action):
         action = int(action)
         reward = 1.0 if action == (self._state % self.num_actions) else 0.0
-        self._state = (self._state * 5 + action + 1) % self.bank_size
+        noise = int(self._rng.integers(4))  # stochastic transitions keep
+        # state coverage broad even under a (near-)deterministic policy
+        self._state = (self._state * 5 + action + 1 + noise) % self.bank_size
         self._steps += 1
         truncated = self._steps >= self.episode_length
         return self.bank[self._state], reward, False, truncated, {}
@@ -90,7 +92,8 @@ class SyntheticAtariVecEnv:
     def step(self, actions: np.ndarray):
         actions = np.asarray(actions, dtype=np.int64)
         reward = (actions == (self._state % self.num_actions)).astype(np.float32)
-        self._state = (self._state * 5 + actions + 1) % self.bank_size
+        noise = self._rng.integers(4, size=self.num_envs)
+        self._state = (self._state * 5 + actions + 1 + noise) % self.bank_size
         self._steps += 1
         done = self._steps >= self.episode_length
         if done.any():

@@ -1,0 +1,104 @@
+"""IMPALA Atari policy network.
+
+Reference semantics (algorithms/utils/atari_model.py:8-143): Nature-CNN
+3-conv encoder (8×8/4→32, 4×4/2→64, 3×3/1→64) + FC 3136→512, core input =
+[features, clipped reward, one-hot last action], optional 2-layer LSTM with
+per-step done masking, policy-logits + baseline heads, multinomial sample in
+train / argmax in eval.  I/O is the [T,B,...] dict wire format.
+
+MI355X design differences from the reference:
+- the LSTM unroll is :class:`scalerl_amd.ops.MaskedLSTM` (hoisted input
+  GEMM + one fused HIP pointwise kernel per step) instead of a Python loop
+  over nn.LSTM;
+- frames stay uint8 until the normalize-on-device divide;
+- the conv/FC stack runs in bf16 under autocast on the learner (heads and
+  LSTM stay fp32).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import MaskedLSTM
+
+
+class AtariNet(nn.Module):
+    def __init__(self, observation_shape=(4, 84, 84), num_actions: int = 6,
+                 use_lstm: bool = True):
+        super().__init__()
+        self.observation_shape = tuple(observation_shape)
+        self.num_actions = num_actions
+        self.use_lstm = use_lstm
+
+        c = observation_shape[0]
+        self.conv1 = nn.Conv2d(c, 32, kernel_size=8, stride=4)
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
+        self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
+        conv_out = 64 * 7 * 7  # 84x84 input
+        self.fc = nn.Linear(conv_out, 512)
+
+        core_output_size = 512 + num_actions + 1
+        if use_lstm:
+            self.core = MaskedLSTM(core_output_size, core_output_size,
+                                   num_layers=2)
+        self.policy = nn.Linear(core_output_size, num_actions)
+        self.baseline = nn.Linear(core_output_size, 1)
+
+    def initial_state(self, batch_size: int, device=None):
+        if not self.use_lstm:
+            return tuple()
+        return self.core.initial_state(batch_size, device=device)
+
+    def encode(self, x: torch.Tensor) -> torch.Tensor:
+        """uint8 [N,C,H,W] → fp feature [N,512]."""
+        x = x.float() / 255.0
+        x = F.relu(self.conv1(x))
+        x = F.relu(self.conv2(x))
+        x = F.relu(self.conv3(x))
+        x = torch.flatten(x, 1)
+        return F.relu(self.fc(x))
+
+    def forward(self, inputs: Dict[str, torch.Tensor],
+                core_state: Tuple = (), greedy: bool = False):
+        """inputs: obs [T,B,C,H,W] u8, reward [T,B], done [T,B] bool,
+        last_action [T,B].  Returns (out dict, new core state)."""
+        obs = inputs["obs"]
+        T, B = obs.shape[:2]
+        feat = self.encode(obs.flatten(0, 1))
+
+        one_hot_action = F.one_hot(
+            inputs["last_action"].view(T * B), self.num_actions).float()
+        clipped_reward = torch.clamp(inputs["reward"].view(T * B, 1), -1, 1)
+        core_input = torch.cat(
+            [feat.float(), clipped_reward, one_hot_action], dim=-1)
+
+        if self.use_lstm:
+            notdone = (~inputs["done"].view(T, B)).float()
+            core_output, core_state = self.core(
+                core_input.view(T, B, -1), notdone, core_state)
+            core_output = core_output.flatten(0, 1)
+        else:
+            core_output = core_input
+
+        # Heads stay fp32 even under autocast: the V-trace ratio
+        # exp(log pi - log mu) is noise-sensitive and behavior logits come
+        # from the fp32 CPU actor.
+        if core_output.is_cuda:
+            with torch.autocast(device_type="cuda", enabled=False):
+                policy_logits = self.policy(core_output.float())
+                baseline = self.baseline(core_output.float())
+        else:
+            policy_logits = self.policy(core_output.float())
+            baseline = self.baseline(core_output.float())
+        if greedy or not self.training:
+            action = torch.argmax(policy_logits, dim=-1)
+        else:
+            action = torch.multinomial(
+                F.softmax(policy_logits, dim=-1), num_samples=1).squeeze(-1)
+        return ({"policy_logits": policy_logits.view(T, B, self.num_actions),
+                 "baseline": baseline.view(T, B),
+                 "action": action.view(T, B)}, core_state)

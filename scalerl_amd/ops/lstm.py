@@ -81,6 +81,7 @@ class _MaskedLSTMFn(torch.autograd.Function):
     """
 
     @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.float32)
     def forward(ctx, x, notdone, h0, c0, w_ih, w_hh, b_ih, b_hh):
         T, B, I = x.shape
         H = w_hh.shape[1]
@@ -108,6 +109,7 @@ class _MaskedLSTMFn(torch.autograd.Function):
         return hs, h, c
 
     @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
     def backward(ctx, d_hs, d_hT, d_cT):
         (x, notdone, hs_in, cs_in, cs_out, gates_all, w_ih,
          w_hh) = ctx.saved_tensors

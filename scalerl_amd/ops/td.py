@@ -47,6 +47,7 @@ def per_is_weights(prios, p_total, p_min, replay_size, beta):
 
 class _FusedTDLossFn(torch.autograd.Function):
     @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.float32)
     def forward(ctx, q, q_next_online, q_next_target, actions, rewards,
                 discounts, prios, p_total, p_min, beta, replay_size, huber,
                 huber_delta):
@@ -74,6 +75,7 @@ class _FusedTDLossFn(torch.autograd.Function):
         return loss_out[0], td_abs
 
     @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
     def backward(ctx, g_loss, g_td):
         (grad_q,) = ctx.saved_tensors
         return (grad_q * g_loss,) + (None,) * 12

@@ -102,6 +102,7 @@ def impala_loss_reference(behavior_logits, target_logits, actions, rewards,
 
 class _ImpalaFusedLossFn(torch.autograd.Function):
     @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.float32)
     def forward(ctx, target_logits, values, behavior_logits, actions, rewards,
                 discounts, bootstrap_value, clip_rho, clip_c, clip_pg_rho,
                 baseline_cost, entropy_cost, want_vs):
@@ -137,6 +138,7 @@ class _ImpalaFusedLossFn(torch.autograd.Function):
         return total, loss_out, loss_out.new_zeros(0)
 
     @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
     def backward(ctx, g_total, g_loss_out, g_vs):
         grad_logits, grad_values = ctx.saved_tensors
         return (grad_logits * g_total, grad_values * g_total, None, None,

@@ -1,0 +1,283 @@
+"""IMPALA rollout transport: shared-memory trajectory slots, actor workers,
+and the learner-side batch gatherer with side-stream H2D copies.
+
+MI355X redesign of the reference's buffer machinery
+(impala_atari.py:122-151 create_buffers, :153-220 get_action actor loop,
+:222-268 get_batch):
+
+- slots are Structure-of-Arrays shared-memory tensors [S, T+1, E, ...] with
+  E envs per actor (the reference is E=1; vectorizing the actor batches its
+  CPU inference and amortizes queue latency);
+- the learner hipHostRegisters the shared region so H2D copies are true
+  async DMA on a dedicated side stream (replaces `.to(device,
+  non_blocking=True)` at impala_atari.py:259-266);
+- free/full index queues carry slot ids (SimpleQueue, as the reference) —
+  latency is amortized over T*E env steps per slot;
+- weight publication: actors' model params alias ONE shared flat fp32
+  buffer (parallel/flat.py); the learner publishes with a single flat copy
+  (replaces load_state_dict at impala_atari.py:348).
+
+Wire format per slot row t in [0, T]:
+  obs[t], reward[t], done[t], last_action[t]  — env output entering step t
+  action[t], logits[t]                        — policy decision taken at t
+  core_state                                  — LSTM (h,c) BEFORE row 0
+Learner consumes rows 0..T-1 of (action, logits), rows 1..T of (reward,
+done), model values on rows 0..T (row T bootstraps).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import ctypes.util
+import os
+import queue
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+
+class RolloutStore:
+    """Shared-memory SoA trajectory slots."""
+
+    def __init__(self, num_slots: int, rollout_length: int, envs_per_slot: int,
+                 obs_shape: Tuple[int, ...], num_actions: int,
+                 lstm_layers: int = 2, lstm_hidden: int = 0):
+        S, T1, E = num_slots, rollout_length + 1, envs_per_slot
+        self.num_slots = S
+        self.rollout_length = rollout_length
+        self.envs_per_slot = E
+        self.num_actions = num_actions
+        self.lstm_layers = lstm_layers
+        self.lstm_hidden = lstm_hidden
+
+        def shared(shape, dtype):
+            t = torch.zeros(shape, dtype=dtype)
+            t.share_memory_()
+            return t
+
+        self.obs = shared((S, T1, E, *obs_shape), torch.uint8)
+        self.reward = shared((S, T1, E), torch.float32)
+        self.done = shared((S, T1, E), torch.bool)
+        self.last_action = shared((S, T1, E), torch.int64)
+        self.action = shared((S, T1, E), torch.int64)
+        self.logits = shared((S, T1, E, num_actions), torch.float32)
+        self.episode_return = shared((S, T1, E), torch.float32)
+        if lstm_hidden > 0:
+            self.core_state = shared((S, 2, lstm_layers, E, lstm_hidden),
+                                     torch.float32)
+        else:
+            self.core_state = None
+
+    def fields(self) -> Dict[str, torch.Tensor]:
+        d = dict(obs=self.obs, reward=self.reward, done=self.done,
+                 last_action=self.last_action, action=self.action,
+                 logits=self.logits, episode_return=self.episode_return)
+        if self.core_state is not None:
+            d["core_state"] = self.core_state
+        return d
+
+    def nbytes(self) -> int:
+        return sum(t.numel() * t.element_size() for t in self.fields().values())
+
+
+def _libhip():
+    try:
+        return ctypes.CDLL("libamdhip64.so")
+    except OSError:
+        return None
+
+
+def pin_store(store: RolloutStore) -> int:
+    """hipHostRegister every shared slot tensor in THIS process so H2D
+    copies from them are async DMA.  Registration is per-process; call it
+    from the learner after CUDA init.  Returns bytes pinned (0 = no-op)."""
+    lib = _libhip()
+    if lib is None or not torch.cuda.is_available():
+        return 0
+    pinned = 0
+    for t in store.fields().values():
+        ptr = t.data_ptr()
+        nbytes = t.numel() * t.element_size()
+        # hipHostRegisterDefault = 0
+        if lib.hipHostRegister(ctypes.c_void_p(ptr), ctypes.c_size_t(nbytes),
+                               ctypes.c_uint(0)) == 0:
+            pinned += nbytes
+    return pinned
+
+
+class ActorState:
+    """Per-actor persistent env/policy state between slots."""
+
+    def __init__(self, env, model, num_envs: int, device="cpu"):
+        self.env = env
+        self.model = model
+        obs = env.reset()
+        E = num_envs
+        self.obs = torch.from_numpy(np.ascontiguousarray(obs))
+        self.reward = torch.zeros(E)
+        self.done = torch.ones(E, dtype=torch.bool)  # episode starts
+        self.last_action = torch.zeros(E, dtype=torch.int64)
+        self.episode_return = torch.zeros(E)
+        if model.use_lstm:
+            self.core_state = model.initial_state(E)
+        else:
+            self.core_state = ()
+
+
+@torch.no_grad()
+def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
+    """Fill one slot (T env steps across E envs).  Returns env steps done."""
+    T = store.rollout_length
+    E = store.envs_per_slot
+    model = state.model
+    if store.core_state is not None:
+        store.core_state[slot][0].copy_(state.core_state[0])
+        store.core_state[slot][1].copy_(state.core_state[1])
+
+    # row 0 = carried-over env output (the continuation row)
+    store.obs[slot, 0].copy_(state.obs)
+    store.reward[slot, 0].copy_(state.reward)
+    store.done[slot, 0].copy_(state.done)
+    store.last_action[slot, 0].copy_(state.last_action)
+    store.episode_return[slot, 0].copy_(state.episode_return)
+
+    for t in range(T):
+        inputs = {
+            "obs": state.obs.unsqueeze(0),
+            "reward": state.reward.unsqueeze(0),
+            "done": state.done.unsqueeze(0),
+            "last_action": state.last_action.unsqueeze(0),
+        }
+        out, state.core_state = model(inputs, state.core_state)
+        action = out["action"].squeeze(0)
+        store.action[slot, t].copy_(action)
+        store.logits[slot, t].copy_(out["policy_logits"].squeeze(0))
+
+        obs_np, rew_np, done_np = state.env.step(action.numpy())
+        # episode accounting: returns BEFORE reset (reward of this step incl.)
+        state.episode_return += torch.from_numpy(rew_np)
+        ep_ret = state.episode_return.clone()
+        done_t = torch.from_numpy(done_np)
+        state.episode_return[done_t] = 0.0
+
+        state.obs = torch.from_numpy(np.ascontiguousarray(obs_np))
+        state.reward = torch.from_numpy(rew_np)
+        state.done = done_t
+        state.last_action = action
+
+        row = t + 1
+        store.obs[slot, row].copy_(state.obs)
+        store.reward[slot, row].copy_(state.reward)
+        store.done[slot, row].copy_(state.done)
+        store.last_action[slot, row].copy_(action)
+        store.episode_return[slot, row].copy_(ep_ret)
+    # row T action/logits: the decision AT the bootstrap row (consumed as
+    # next slot's row-0 last_action; learner ignores it)
+    inputs = {
+        "obs": state.obs.unsqueeze(0),
+        "reward": state.reward.unsqueeze(0),
+        "done": state.done.unsqueeze(0),
+        "last_action": state.last_action.unsqueeze(0),
+    }
+    out, _ = model(inputs, state.core_state)
+    store.action[slot, T].copy_(out["action"].squeeze(0))
+    store.logits[slot, T].copy_(out["policy_logits"].squeeze(0))
+    return T * E
+
+
+def actor_loop(actor_id: int, make_env_fn, model, store: RolloutStore,
+               free_q, full_q, stop_event, step_counter,
+               episode_queue=None, seed: int = 0, torch_threads: int = 1):
+    """Actor process main (reference: impala_atari.py:153-220).
+
+    `model`'s parameters alias the learner-published shared flat buffer;
+    inference here is CPU fp32.  Poison pill: a None on free_q exits.
+    """
+    torch.manual_seed(seed + actor_id)
+    torch.set_num_threads(torch_threads)
+    model.eval()
+    env = make_env_fn(actor_id)
+    state = ActorState(env, model, store.envs_per_slot)
+    try:
+        while not stop_event.is_set():
+            slot = free_q.get()
+            if slot is None:
+                break
+            steps = run_rollout(state, store, slot)
+            full_q.put(slot)
+            with step_counter.get_lock():
+                step_counter.value += steps
+            if episode_queue is not None:
+                # completed-episode returns for logging (best effort)
+                mask = store.done[slot, 1:].numpy()
+                if mask.any():
+                    rets = store.episode_return[slot, 1:].numpy()[mask]
+                    try:
+                        episode_queue.put_nowait(rets.tolist())
+                    except queue.Full:
+                        pass
+    except KeyboardInterrupt:
+        pass
+
+
+class BatchGatherer:
+    """Learner-side: gather K slots into device batch tensors [T+1, K*E, ...]
+    via side-stream async H2D copies (the device boundary of the pipeline,
+    replacing impala_atari.py:248-266)."""
+
+    def __init__(self, store: RolloutStore, device: torch.device,
+                 slots_per_batch: int):
+        self.store = store
+        self.device = device
+        self.K = slots_per_batch
+        self.is_cuda = device.type == "cuda"
+        T1 = store.rollout_length + 1
+        E = store.envs_per_slot
+        K = self.K
+        if self.is_cuda:
+            self.stream = torch.cuda.Stream(device=device)
+            self.staging = {
+                name: torch.empty((K, *t.shape[1:]), dtype=t.dtype,
+                                  device=device)
+                for name, t in store.fields().items()
+            }
+            self.ready_event = torch.cuda.Event()
+
+    def gather(self, slot_ids: List[int]) -> Dict[str, torch.Tensor]:
+        """Copy slots to device; returns batch dict with [T+1, B, ...] layout
+        (B = K*E).  Caller may recycle the slots after `copies_done()`."""
+        store = self.store
+        if not self.is_cuda:
+            batch = {}
+            for name, t in store.fields().items():
+                stacked = torch.stack([t[s] for s in slot_ids])  # [K, ...]
+                batch[name] = self._to_batch_layout(name, stacked)
+            return batch
+        with torch.cuda.stream(self.stream):
+            for name, t in store.fields().items():
+                dst = self.staging[name]
+                for j, s in enumerate(slot_ids):
+                    dst[j].copy_(t[s], non_blocking=True)
+            self.ready_event.record(self.stream)
+        torch.cuda.current_stream(self.device).wait_event(self.ready_event)
+        return {name: self._to_batch_layout(name, self.staging[name])
+                for name in self.staging}
+
+    def copies_done(self) -> None:
+        """Block host until the H2D copies finished (slots reusable)."""
+        if self.is_cuda:
+            self.ready_event.synchronize()
+
+    def _to_batch_layout(self, name: str, stacked: torch.Tensor) -> torch.Tensor:
+        """[K, T+1, E, ...] → [T+1, K*E, ...]; core_state → [L, K*E, H] x2."""
+        K = stacked.shape[0]
+        if name == "core_state":
+            # [K, 2, L, E, H] → [2, L, K*E, H]
+            _, _, L, E, H = stacked.shape
+            return stacked.permute(1, 2, 0, 3, 4).reshape(2, L, K * E, H).contiguous()
+        T1, E = stacked.shape[1], stacked.shape[2]
+        rest = stacked.shape[3:]
+        return stacked.permute(1, 0, 2, *range(3, stacked.dim())).reshape(
+            T1, K * E, *rest).contiguous()

@@ -1,0 +1,306 @@
+"""IMPALA on the MI355X actor-learner runtime — the flagship trainer
+(benchmark config 3: synthetic Atari, 1/2/4/8 learner GPUs).
+
+Reference architecture being reimplemented (impala_atari.py:40-521):
+actor procs fill shared rollout slots via free/full queues; learner batches
+slots, runs V-trace + losses, RMSProp, publishes weights back.  MI355X
+design:
+
+- per-rank topology: each learner rank (1 process per GPU) owns its own
+  actor processes + rollout store; ranks sync ONLY via one flat-grad
+  all-reduce per learn step (weak scaling over RCCL/xGMI);
+- learner hot path: model fwd (bf16 convs) → fused HIP V-trace+loss kernel
+  → backward → flat-grad clip (HIP) → RCCL all-reduce → fused RMSProp (HIP);
+- weight publication: one D2H flat copy into the shared CPU flat buffer the
+  actor models alias (impala_atari.py:348 equivalent);
+- process discipline: actors are forked BEFORE any HIP/CUDA init (fork
+  after GPU init is unsafe on ROCm, and actors never touch the GPU).
+
+Step accounting matches the reference (impala_atari.py:391): one learn
+iteration consumes rollout_length × batch_size env steps.
+"""
+
+from __future__ import annotations
+
+import copy
+import os
+import time
+from typing import Dict, List, Optional
+
+import torch
+import torch.multiprocessing as mp
+
+from ..config import ImpalaArguments
+from ..envs.synthetic import SyntheticAtariVecEnv
+from ..models.atari import AtariNet
+from ..ops import FusedRMSprop, clip_grad_norm_, impala_loss
+from ..parallel import FlatParams, all_reduce_flat, get_rank, get_world_size
+from ..parallel.rollout import (ActorState, BatchGatherer, RolloutStore,
+                                actor_loop, pin_store)
+from ..utils import Timings, get_logger
+from ..utils.checkpoint import load_checkpoint, save_checkpoint
+
+
+def _make_actor_env_factory(args: ImpalaArguments):
+    env_id = args.env_id
+    E = args.envs_per_actor
+
+    def factory(actor_id: int):
+        if env_id == "synthetic-atari":
+            return SyntheticAtariVecEnv(E, seed=args.seed * 1000 + actor_id)
+        # real envs: vectorize single envs with the DeepMind stack
+        from ..envs.vec_env import SyncVectorEnv
+        from ..envs.registry import make_env
+
+        def one(i):
+            return make_env(env_id, seed=args.seed * 1000 + actor_id * E + i,
+                            deepmind_wrap=True)
+        return SyncVectorEnv([lambda i=i: one(i) for i in range(E)])
+    return factory
+
+
+class ImpalaTrainer:
+    def __init__(self, args: ImpalaArguments, device: Optional[str] = None):
+        self.args = args
+        self.rank = get_rank()
+        self.world_size = get_world_size()
+        self.log = get_logger("impala")
+        if device is None:
+            device = ("cuda" if torch.cuda.is_available() else "cpu") \
+                if args.device == "auto" else args.device
+        self.device = torch.device(device)
+
+        # ---- shapes ----
+        probe_env = SyntheticAtariVecEnv(1) if args.env_id == "synthetic-atari" \
+            else None
+        if probe_env is not None:
+            self.obs_shape = probe_env.observation_space.shape
+            self.num_actions = probe_env.action_space.n
+        else:
+            from ..envs.registry import make_env
+            e = make_env(args.env_id, deepmind_wrap=True)
+            self.obs_shape = e.observation_space.shape
+            self.num_actions = e.action_space.n
+            e.close()
+
+        E = args.envs_per_actor
+        assert args.batch_size % E == 0, \
+            f"batch_size ({args.batch_size}) must be a multiple of " \
+            f"envs_per_actor ({E})"
+        self.slots_per_batch = args.batch_size // E
+        num_buffers = args.num_buffers or (
+            2 * args.num_actors + 2 * self.slots_per_batch)
+
+        # ---- CPU phase: shared actor model + store + actor processes ----
+        # (everything here must precede any CUDA/HIP initialization)
+        torch.manual_seed(args.seed + self.rank)
+        self.actor_model = AtariNet(self.obs_shape, self.num_actions,
+                                    use_lstm=args.use_lstm)
+        self.actor_model.eval()
+        self.shared_flat = FlatParams(self.actor_model, device="cpu",
+                                      share=True)
+        lstm_hidden = (512 + self.num_actions + 1) if args.use_lstm else 0
+        self.store = RolloutStore(
+            num_buffers, args.rollout_length, E, self.obs_shape,
+            self.num_actions, lstm_layers=2, lstm_hidden=lstm_hidden)
+
+        ctx = mp.get_context("fork")
+        self.free_q = ctx.SimpleQueue()
+        self.full_q = ctx.SimpleQueue()
+        self.stop_event = ctx.Event()
+        self.step_counter = ctx.Value("l", 0)
+        self.episode_q = ctx.Queue(maxsize=256)
+        self.actors: List[mp.Process] = []
+        self._started = False
+        self._num_buffers = num_buffers
+
+        # ---- learner state (device init deferred to setup()) ----
+        self.learner_model: Optional[AtariNet] = None
+        self.global_step = 0
+        self.learn_iters = 0
+        self.timings = Timings()
+
+    # -- lifecycle ---------------------------------------------------------
+    def start_actors(self) -> None:
+        if self._started:
+            return
+        factory = _make_actor_env_factory(self.args)
+        ctx = mp.get_context("fork")
+        for i in range(self.args.num_actors):
+            p = ctx.Process(
+                target=actor_loop,
+                args=(i, factory, self.actor_model, self.store, self.free_q,
+                      self.full_q, self.stop_event, self.step_counter,
+                      self.episode_q, self.args.seed),
+                daemon=True, name=f"impala-actor-{self.rank}-{i}")
+            p.start()
+            self.actors.append(p)
+        for s in range(self._num_buffers):
+            self.free_q.put(s)
+        self._started = True
+
+    def setup_learner(self) -> None:
+        """Device-side init (safe to call after start_actors)."""
+        args = self.args
+        self.learner_model = AtariNet(self.obs_shape, self.num_actions,
+                                      use_lstm=args.use_lstm).to(self.device)
+        # identical init across ranks and with the actor model
+        self.learner_model.load_state_dict(self.actor_model.state_dict())
+        self.flat = FlatParams(self.learner_model, device=self.device)
+        self.optimizer = FusedRMSprop(
+            self.flat.flat, lr=args.learning_rate, alpha=args.rmsprop_alpha,
+            eps=args.rmsprop_eps, momentum=args.rmsprop_momentum)
+        self.gatherer = BatchGatherer(self.store, self.device,
+                                      self.slots_per_batch)
+        if self.device.type == "cuda":
+            pinned = pin_store(self.store)
+            self.log.info(f"pinned {pinned/1e6:.1f} MB of rollout store")
+            self.publish_stream = torch.cuda.Stream(device=self.device)
+        self._publish_weights()
+        self.autocast_dtype = (torch.bfloat16 if args.dtype == "bf16" and
+                               self.device.type == "cuda" else None)
+
+    @torch.no_grad()
+    def _publish_weights(self) -> None:
+        """learner flat → shared CPU flat (one memcpy; actors alias it)."""
+        if self.device.type == "cuda":
+            with torch.cuda.stream(self.publish_stream):
+                self.shared_flat.flat.copy_(self.flat.flat, non_blocking=True)
+        else:
+            self.shared_flat.flat.copy_(self.flat.flat)
+
+    # -- core step ---------------------------------------------------------
+    def next_batch(self) -> Dict[str, torch.Tensor]:
+        slot_ids = [self.full_q.get() for _ in range(self.slots_per_batch)]
+        self.timings.time("dequeue")
+        batch = self.gatherer.gather(slot_ids)
+        self.timings.time("gather")
+        self.gatherer.copies_done()
+        for s in slot_ids:
+            self.free_q.put(s)
+        self.timings.time("enqueue")
+        return batch
+
+    def learn_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        args = self.args
+        model = self.learner_model
+        model.train()
+        self.flat.flat_grad.zero_()
+
+        inputs = {"obs": batch["obs"], "reward": batch["reward"],
+                  "done": batch["done"], "last_action": batch["last_action"]}
+        core_state = ()
+        if args.use_lstm:
+            cs = batch["core_state"]
+            core_state = (cs[0], cs[1])
+        if self.autocast_dtype is not None:
+            with torch.autocast(device_type="cuda", dtype=self.autocast_dtype):
+                out, _ = model(inputs, core_state, greedy=True)
+        else:
+            out, _ = model(inputs, core_state, greedy=True)
+        logits = out["policy_logits"].float()
+        baseline = out["baseline"].float()
+
+        rewards = batch["reward"][1:]
+        if args.reward_clipping == "abs_one":
+            rewards = torch.clamp(rewards, -1, 1)
+        discounts = (~batch["done"][1:]).float() * args.discounting
+
+        total, comps, _ = impala_loss(
+            batch["logits"][:-1], logits[:-1], batch["action"][:-1],
+            rewards, discounts, baseline[:-1], baseline[-1].detach(),
+            clip_rho_threshold=args.clip_rho_threshold,
+            clip_c_threshold=1.0,
+            clip_pg_rho_threshold=args.clip_pg_rho_threshold,
+            baseline_cost=args.baseline_cost, entropy_cost=args.entropy_cost)
+        self.timings.time("forward")
+        total.backward()
+        self.timings.time("backward")
+
+        all_reduce_flat(self.flat.flat_grad, average=True)
+        clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
+        self.optimizer.step(self.flat.flat_grad)
+        self.timings.time("optimize")
+        self._publish_weights()
+        self.timings.time("publish")
+
+        self.learn_iters += 1
+        self.global_step += args.rollout_length * args.batch_size
+        return {"total_loss": float(total.detach()),
+                "pg_loss": float(comps[0]),
+                "baseline_loss": float(comps[1]),
+                "entropy_loss": float(comps[2])}
+
+    def train_iteration(self) -> Dict[str, float]:
+        self.timings.reset()
+        batch = self.next_batch()
+        return self.learn_step(batch)
+
+    # -- driver ------------------------------------------------------------
+    def train(self) -> None:
+        args = self.args
+        self.start_actors()
+        self.setup_learner()
+        ckpt_path = args.checkpoint_path or os.path.join(
+            args.output_dir, "model.tar")
+        last_ckpt = time.time()
+        last_log = time.time()
+        last_step = 0
+        stats: Dict[str, float] = {}
+        try:
+            while self.global_step < args.total_steps:
+                stats = self.train_iteration()
+                now = time.time()
+                if self.rank == 0 and now - last_log > 5.0:
+                    sps = (self.global_step - last_step) / (now - last_log)
+                    last_log, last_step = now, self.global_step
+                    rets = self._drain_episode_returns()
+                    ret_str = (f" ret={sum(rets)/len(rets):.2f}" if rets else "")
+                    self.log.info(
+                        f"step {self.global_step} SPS {sps:,.0f} "
+                        f"loss {stats['total_loss']:.3f}{ret_str}")
+                if (self.rank == 0 and not args.disable_checkpoint and
+                        now - last_ckpt > args.checkpoint_interval_s):
+                    self.save(ckpt_path)
+                    last_ckpt = now
+        finally:
+            if self.rank == 0 and not args.disable_checkpoint:
+                self.save(ckpt_path)
+            self.shutdown()
+
+    def _drain_episode_returns(self) -> List[float]:
+        rets: List[float] = []
+        try:
+            while True:
+                rets.extend(self.episode_q.get_nowait())
+        except Exception:
+            pass
+        return rets
+
+    def save(self, path: str) -> None:
+        """IMPALA-format checkpoint (model.tar keys, impala_atari.py:503)."""
+        save_checkpoint(path, model=self.learner_model,
+                        optimizer=self.optimizer,
+                        hparam=vars(self.args),
+                        extra={"global_step": self.global_step})
+
+    def load(self, path: str) -> None:
+        ckpt = load_checkpoint(path, map_location=self.device)
+        self.learner_model.load_state_dict(ckpt["model_state_dict"])
+        # load_state_dict replaced param storages; re-flatten
+        self.flat = FlatParams(self.learner_model, device=self.device)
+        opt_sd = ckpt.get("optimizer_state_dict")
+        if opt_sd:
+            self.optimizer.load_state_dict(opt_sd)
+        self.global_step = int(ckpt.get("global_step", 0))
+        self._publish_weights()
+
+    def shutdown(self) -> None:
+        self.stop_event.set()
+        for _ in self.actors:
+            self.free_q.put(None)  # poison pills (impala_atari.py:480)
+        for p in self.actors:
+            p.join(timeout=2.0)
+            if p.is_alive():
+                p.terminate()
+        self.actors.clear()

@@ -1,0 +1,86 @@
+"""End-to-end IMPALA on CPU: actor procs → shared slots → learner →
+checkpoint round trip; plus a learning-signal check on the synthetic env."""
+
+import os
+
+import pytest
+import torch
+
+from scalerl_amd.config import ImpalaArguments
+from scalerl_amd.runtime.impala import ImpalaTrainer
+
+
+def _args(tmp_path, **kw):
+    base = dict(rollout_length=8, batch_size=8, envs_per_actor=4,
+                num_actors=2, total_steps=8 * 8 * 4, use_lstm=True,
+                device="cpu", dtype="fp32", output_dir=str(tmp_path),
+                checkpoint_interval_s=1e9, seed=7)
+    base.update(kw)
+    return ImpalaArguments(**base)
+
+
+def test_impala_cpu_end_to_end(tmp_path):
+    t = ImpalaTrainer(_args(tmp_path))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        losses = [t.train_iteration()["total_loss"] for _ in range(4)]
+        assert all(torch.isfinite(torch.tensor(losses)))
+        assert t.global_step == 4 * 8 * 8
+        # checkpoint round trip (IMPALA model.tar format)
+        path = os.path.join(str(tmp_path), "model.tar")
+        t.save(path)
+        ckpt = torch.load(path, map_location="cpu", weights_only=False)
+        assert set(ckpt) >= {"model_state_dict", "optimizer_state_dict",
+                             "hparam"}
+        before = t.flat.flat.clone()
+        t.flat.flat.add_(1.0)  # corrupt
+        t.load(path)
+        torch.testing.assert_close(t.flat.flat, before)
+    finally:
+        t.shutdown()
+
+
+def test_impala_weight_publication_reaches_actors(tmp_path):
+    t = ImpalaTrainer(_args(tmp_path))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        t.train_iteration()
+        # after publish, shared CPU flat == learner flat
+        torch.testing.assert_close(t.shared_flat.flat, t.flat.flat.cpu())
+    finally:
+        t.shutdown()
+
+
+def test_impala_no_lstm_path(tmp_path):
+    t = ImpalaTrainer(_args(tmp_path, use_lstm=False))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        stats = t.train_iteration()
+        assert torch.isfinite(torch.tensor(stats["total_loss"]))
+    finally:
+        t.shutdown()
+
+
+def test_impala_learns_synthetic_reward(tmp_path):
+    """The synthetic env rewards action == state % A; 120 iterations of
+    IMPALA should raise average reward above the uniform-policy baseline
+    (1/6 ≈ 0.167)."""
+    t = ImpalaTrainer(_args(tmp_path, rollout_length=16, batch_size=16,
+                            envs_per_actor=8, num_actors=2, use_lstm=False,
+                            entropy_cost=0.02, learning_rate=3e-4,
+                            discounting=0.5))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        rews = []
+        for i in range(120):
+            t.train_iteration()
+            if i >= 100:
+                rews.append(t.next_batch()["reward"].mean().item())
+        avg = sum(rews) / len(rews)
+        assert avg > 0.22, f"no learning signal: avg reward {avg:.3f}"
+    finally:
+        t.shutdown()
