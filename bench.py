@@ -52,17 +52,23 @@ def main():
     use_cuda = torch.cuda.is_available() and args.device in ("auto", "cuda")
     device = f"cuda:{local_rank}" if use_cuda else "cpu"
 
-    if args.num_actors <= 0:
-        ncpu = os.cpu_count() or 8
-        # leave ~2 cores per rank for the learner process + OS
-        args.num_actors = max(2, (ncpu - 2 * world) // max(1, world))
-
     from scalerl_amd.config import ImpalaArguments
     from scalerl_amd.runtime.impala import ImpalaTrainer
 
     inference = args.inference
     if inference == "auto":
         inference = "gpu" if use_cuda else "cpu"
+
+    if args.num_actors <= 0:
+        ncpu = os.cpu_count() or 8
+        avail = max(2, (ncpu - 3 * world) // max(1, world))
+        if inference == "gpu":
+            # actors are pure env-steppers; a handful saturate the
+            # inference worker
+            args.num_actors = min(16, avail)
+        else:
+            # one CPU-inference actor ≈ a few hundred steps/s
+            args.num_actors = min(64, avail)
 
     cfg = ImpalaArguments(
         rollout_length=args.rollout_length, batch_size=args.batch_size,
@@ -111,6 +117,13 @@ def main():
     value = total_env_steps / elapsed
 
     if rank == 0:
+        import sys
+        produced = trainer.step_counter.value
+        print(f"[bench] learner timings (s/iter means):\n"
+              f"{trainer.timings.summary('  ')}\n"
+              f"[bench] actor production during run: {produced} env-steps "
+              f"({produced / elapsed:,.0f}/s incl. warmup overlap)",
+              file=sys.stderr)
         result = {
             "metric": "env_steps_per_sec",
             "value": round(value, 1),
@@ -132,7 +145,7 @@ def main():
                 "seq_len": args.rollout_length,
                 "envs_per_actor": args.envs_per_actor,
                 "actors_per_rank": args.num_actors,
-                "inference": inference,
+                "inference": trainer.inference,  # actual placement used
                 "parallelism": f"dp{world}",
             },
         }

@@ -1,0 +1,197 @@
+"""Centralized GPU inference for actors (SEED-RL-style).
+
+The reference runs per-actor CPU inference against a shared-memory model
+(impala_atari.py:196-198).  That burns one CPU core per ~500 env-steps/s;
+on an 8-GPU node the cores run out long before the learners do.  The
+MI355X-native alternative: actor processes are pure env-steppers, and each
+learner rank runs ONE inference worker process that batches all its actors'
+observation batches through the policy on the rank's GPU:
+
+- per-actor request/response slots in shared memory (hipHostRegistered by
+  the worker → async H2D/D2H DMA);
+- request fan-in over an mp.Queue; responses signalled per-actor via
+  semaphores;
+- the policy's LSTM states for every actor env live on the GPU inside the
+  worker; actors never see them.  At rollout-slot start the actor sets
+  `want_state`, and the worker writes the pre-step (h,c) snapshot back so
+  the learner can initialize its unroll (replaces the reference's
+  create_rnn_state_buffers, impala_atari.py:108-120);
+- weights refresh from the learner-published shared CPU flat buffer
+  (parallel/flat.py), gated by a version counter — one H2D flat copy.
+
+Both processes share the GPU: inference forwards interleave with learner
+kernels (separate HIP contexts timeslice the CU array).
+"""
+
+from __future__ import annotations
+
+import ctypes
+import queue
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.multiprocessing as mp
+
+
+class InferenceSlots:
+    """Shared-memory request/response slots, one row per actor."""
+
+    def __init__(self, num_actors: int, envs_per_actor: int,
+                 obs_shape: Tuple[int, ...], num_actions: int,
+                 lstm_layers: int, lstm_hidden: int):
+        A, E = num_actors, envs_per_actor
+        self.num_actors, self.envs_per_actor = A, E
+        self.num_actions = num_actions
+        self.lstm_layers, self.lstm_hidden = lstm_layers, lstm_hidden
+
+        def shared(shape, dtype):
+            t = torch.zeros(shape, dtype=dtype)
+            t.share_memory_()
+            return t
+
+        self.obs = shared((A, E, *obs_shape), torch.uint8)
+        self.reward = shared((A, E), torch.float32)
+        self.done = shared((A, E), torch.bool)
+        self.last_action = shared((A, E), torch.int64)
+        self.want_state = shared((A,), torch.int32)
+        self.action = shared((A, E), torch.int64)
+        self.logits = shared((A, E, num_actions), torch.float32)
+        if lstm_hidden > 0:
+            self.core_state = shared((A, 2, lstm_layers, E, lstm_hidden),
+                                     torch.float32)
+        else:
+            self.core_state = None
+
+    def tensors(self):
+        ts = [self.obs, self.reward, self.done, self.last_action,
+              self.want_state, self.action, self.logits]
+        if self.core_state is not None:
+            ts.append(self.core_state)
+        return ts
+
+
+class RemotePolicy:
+    """Actor-side client: request an inference round from the worker."""
+
+    def __init__(self, actor_id: int, slots: InferenceSlots, req_q, sem):
+        self.aid = actor_id
+        self.slots = slots
+        self.req_q = req_q
+        self.sem = sem
+
+    def __call__(self, obs, reward, done, last_action,
+                 want_state: bool = False):
+        a = self.aid
+        s = self.slots
+        s.obs[a].copy_(obs)
+        s.reward[a].copy_(reward)
+        s.done[a].copy_(done)
+        s.last_action[a].copy_(last_action)
+        s.want_state[a] = 1 if want_state else 0
+        self.req_q.put(a)
+        self.sem.acquire()
+        state = s.core_state[a].clone() if (want_state and
+                                            s.core_state is not None) else None
+        return s.action[a].clone(), s.logits[a].clone(), state
+
+
+def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlots,
+                     shared_flat: torch.Tensor, version: "mp.Value",
+                     req_q, sems: List, stop_event, max_batch_actors: int = 0,
+                     seed: int = 0):
+    """Inference worker process main."""
+    from ..models.atari import AtariNet
+    from .flat import FlatParams
+    from .rollout import pin_tensor
+
+    torch.manual_seed(seed)
+    device = torch.device(f"cuda:{device_index}")
+    torch.cuda.set_device(device)
+    model = AtariNet(**model_kwargs).to(device)
+    model.train()  # multinomial action sampling
+    flat = FlatParams(model, device=device)
+    for t in slots.tensors():
+        pin_tensor(t)
+    pin_tensor(shared_flat)
+
+    A, E = slots.num_actors, slots.envs_per_actor
+    nact = slots.num_actions
+    use_lstm = slots.core_state is not None
+    if use_lstm:
+        L, H = slots.lstm_layers, slots.lstm_hidden
+        h_all = torch.zeros(L, A * E, H, device=device)
+        c_all = torch.zeros(L, A * E, H, device=device)
+    max_batch_actors = max_batch_actors or A
+    seen_version = -1
+
+    # device staging for a batched round
+    obs_d = torch.empty((A, E, *slots.obs.shape[2:]), dtype=torch.uint8,
+                        device=device)
+    rew_d = torch.empty((A, E), device=device)
+    done_d = torch.empty((A, E), dtype=torch.bool, device=device)
+    lastact_d = torch.empty((A, E), dtype=torch.int64, device=device)
+
+    with torch.no_grad():
+        while not stop_event.is_set():
+            try:
+                first = req_q.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            if first is None:
+                break
+            ids = [first]
+            while len(ids) < max_batch_actors:
+                try:
+                    nxt = req_q.get_nowait()
+                except queue.Empty:
+                    break
+                if nxt is None:
+                    stop_event.set()
+                    break
+                ids.append(nxt)
+            if version.value != seen_version:
+                seen_version = version.value
+                flat.load_from(shared_flat, non_blocking=False)
+
+            k = len(ids)
+            # H2D for the requesting actors
+            for j, a in enumerate(ids):
+                obs_d[j].copy_(slots.obs[a], non_blocking=True)
+                rew_d[j].copy_(slots.reward[a], non_blocking=True)
+                done_d[j].copy_(slots.done[a], non_blocking=True)
+                lastact_d[j].copy_(slots.last_action[a], non_blocking=True)
+            cols = torch.tensor([a * E + e for a in ids for e in range(E)],
+                                device=device)
+            inputs = {
+                "obs": obs_d[:k].reshape(1, k * E, *slots.obs.shape[2:]),
+                "reward": rew_d[:k].reshape(1, k * E),
+                "done": done_d[:k].reshape(1, k * E),
+                "last_action": lastact_d[:k].reshape(1, k * E),
+            }
+            state = ()
+            snap = {}
+            if use_lstm:
+                h = h_all.index_select(1, cols)
+                c = c_all.index_select(1, cols)
+                for j, a in enumerate(ids):
+                    if slots.want_state[a]:
+                        snap[a] = (j, h[:, j * E:(j + 1) * E].clone(),
+                                   c[:, j * E:(j + 1) * E].clone())
+                state = (h, c)
+            out, new_state = model(inputs, state)
+            if use_lstm:
+                h_all.index_copy_(1, cols, new_state[0])
+                c_all.index_copy_(1, cols, new_state[1])
+            action = out["action"].view(k, E)
+            logits = out["policy_logits"].view(k, E, nact)
+            for j, a in enumerate(ids):
+                slots.action[a].copy_(action[j], non_blocking=True)
+                slots.logits[a].copy_(logits[j], non_blocking=True)
+                if a in snap:
+                    _, hj, cj = snap[a]
+                    slots.core_state[a][0].copy_(hj, non_blocking=True)
+                    slots.core_state[a][1].copy_(cj, non_blocking=True)
+            torch.cuda.synchronize()
+            for a in ids:
+                sems[a].release()

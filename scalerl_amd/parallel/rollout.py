@@ -89,6 +89,18 @@ def _libhip():
         return None
 
 
+def pin_tensor(t: torch.Tensor) -> int:
+    """hipHostRegister one CPU tensor in this process (0 on failure)."""
+    lib = _libhip()
+    if lib is None or not torch.cuda.is_available():
+        return 0
+    nbytes = t.numel() * t.element_size()
+    if lib.hipHostRegister(ctypes.c_void_p(t.data_ptr()),
+                           ctypes.c_size_t(nbytes), ctypes.c_uint(0)) == 0:
+        return nbytes
+    return 0
+
+
 def pin_store(store: RolloutStore) -> int:
     """hipHostRegister every shared slot tensor in THIS process so H2D
     copies from them are async DMA.  Registration is per-process; call it
@@ -107,12 +119,37 @@ def pin_store(store: RolloutStore) -> int:
     return pinned
 
 
-class ActorState:
-    """Per-actor persistent env/policy state between slots."""
+class LocalPolicy:
+    """CPU inference against the shared-flat-aliased model (the reference's
+    per-actor inference mode, impala_atari.py:196-198).  Same call contract
+    as :class:`scalerl_amd.parallel.inference.RemotePolicy`."""
 
-    def __init__(self, env, model, num_envs: int, device="cpu"):
-        self.env = env
+    def __init__(self, model, num_envs: int):
         self.model = model
+        if model.use_lstm:
+            self.core_state = model.initial_state(num_envs)
+        else:
+            self.core_state = ()
+
+    @torch.no_grad()
+    def __call__(self, obs, reward, done, last_action, want_state=False):
+        snap = None
+        if want_state and self.model.use_lstm:
+            snap = torch.stack([self.core_state[0], self.core_state[1]])
+        inputs = {"obs": obs.unsqueeze(0), "reward": reward.unsqueeze(0),
+                  "done": done.unsqueeze(0),
+                  "last_action": last_action.unsqueeze(0)}
+        out, self.core_state = self.model(inputs, self.core_state)
+        return (out["action"].squeeze(0), out["policy_logits"].squeeze(0),
+                snap)
+
+
+class ActorState:
+    """Per-actor persistent env state between slots."""
+
+    def __init__(self, env, policy, num_envs: int):
+        self.env = env
+        self.policy = policy
         obs = env.reset()
         E = num_envs
         self.obs = torch.from_numpy(np.ascontiguousarray(obs))
@@ -120,21 +157,18 @@ class ActorState:
         self.done = torch.ones(E, dtype=torch.bool)  # episode starts
         self.last_action = torch.zeros(E, dtype=torch.int64)
         self.episode_return = torch.zeros(E)
-        if model.use_lstm:
-            self.core_state = model.initial_state(E)
-        else:
-            self.core_state = ()
 
 
 @torch.no_grad()
 def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
-    """Fill one slot (T env steps across E envs).  Returns env steps done."""
+    """Fill one slot (T env steps across E envs).  Returns env steps done.
+
+    Row layout (see module docstring): action[t]/logits[t] = decision taken
+    at the state in row t; rows written 0..T for env fields, 0..T-1 for
+    decisions (row T's decision belongs to the next slot's row 0).
+    """
     T = store.rollout_length
     E = store.envs_per_slot
-    model = state.model
-    if store.core_state is not None:
-        store.core_state[slot][0].copy_(state.core_state[0])
-        store.core_state[slot][1].copy_(state.core_state[1])
 
     # row 0 = carried-over env output (the continuation row)
     store.obs[slot, 0].copy_(state.obs)
@@ -144,16 +178,13 @@ def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
     store.episode_return[slot, 0].copy_(state.episode_return)
 
     for t in range(T):
-        inputs = {
-            "obs": state.obs.unsqueeze(0),
-            "reward": state.reward.unsqueeze(0),
-            "done": state.done.unsqueeze(0),
-            "last_action": state.last_action.unsqueeze(0),
-        }
-        out, state.core_state = model(inputs, state.core_state)
-        action = out["action"].squeeze(0)
+        action, logits, snap = state.policy(
+            state.obs, state.reward, state.done, state.last_action,
+            want_state=(t == 0 and store.core_state is not None))
+        if snap is not None:
+            store.core_state[slot].copy_(snap)
         store.action[slot, t].copy_(action)
-        store.logits[slot, t].copy_(out["policy_logits"].squeeze(0))
+        store.logits[slot, t].copy_(logits)
 
         obs_np, rew_np, done_np = state.env.step(action.numpy())
         # episode accounting: returns BEFORE reset (reward of this step incl.)
@@ -173,33 +204,23 @@ def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
         store.done[slot, row].copy_(state.done)
         store.last_action[slot, row].copy_(action)
         store.episode_return[slot, row].copy_(ep_ret)
-    # row T action/logits: the decision AT the bootstrap row (consumed as
-    # next slot's row-0 last_action; learner ignores it)
-    inputs = {
-        "obs": state.obs.unsqueeze(0),
-        "reward": state.reward.unsqueeze(0),
-        "done": state.done.unsqueeze(0),
-        "last_action": state.last_action.unsqueeze(0),
-    }
-    out, _ = model(inputs, state.core_state)
-    store.action[slot, T].copy_(out["action"].squeeze(0))
-    store.logits[slot, T].copy_(out["policy_logits"].squeeze(0))
     return T * E
 
 
-def actor_loop(actor_id: int, make_env_fn, model, store: RolloutStore,
+def actor_loop(actor_id: int, make_env_fn, make_policy_fn, store: RolloutStore,
                free_q, full_q, stop_event, step_counter,
                episode_queue=None, seed: int = 0, torch_threads: int = 1):
     """Actor process main (reference: impala_atari.py:153-220).
 
-    `model`'s parameters alias the learner-published shared flat buffer;
-    inference here is CPU fp32.  Poison pill: a None on free_q exits.
+    ``make_policy_fn(actor_id)`` returns either a LocalPolicy (CPU shared
+    model) or a RemotePolicy (GPU inference worker client).  Poison pill: a
+    None on free_q exits.
     """
     torch.manual_seed(seed + actor_id)
     torch.set_num_threads(torch_threads)
-    model.eval()
     env = make_env_fn(actor_id)
-    state = ActorState(env, model, store.envs_per_slot)
+    policy = make_policy_fn(actor_id)
+    state = ActorState(env, policy, store.envs_per_slot)
     try:
         while not stop_event.is_set():
             slot = free_q.get()

@@ -35,8 +35,10 @@ from ..envs.synthetic import SyntheticAtariVecEnv
 from ..models.atari import AtariNet
 from ..ops import FusedRMSprop, clip_grad_norm_, impala_loss
 from ..parallel import FlatParams, all_reduce_flat, get_rank, get_world_size
-from ..parallel.rollout import (ActorState, BatchGatherer, RolloutStore,
-                                actor_loop, pin_store)
+from ..parallel.inference import (InferenceSlots, RemotePolicy,
+                                  inference_worker)
+from ..parallel.rollout import (ActorState, BatchGatherer, LocalPolicy,
+                                RolloutStore, actor_loop, pin_store)
 from ..utils import Timings, get_logger
 from ..utils.checkpoint import load_checkpoint, save_checkpoint
 
@@ -114,6 +116,19 @@ class ImpalaTrainer:
         self._started = False
         self._num_buffers = num_buffers
 
+        # ---- inference placement ----
+        self.inference = args.inference
+        if self.inference == "gpu" and self.device.type != "cuda":
+            self.inference = "cpu"
+        self.inference_proc: Optional[mp.Process] = None
+        if self.inference == "gpu":
+            self.inf_slots = InferenceSlots(
+                args.num_actors, E, self.obs_shape, self.num_actions,
+                lstm_layers=2, lstm_hidden=lstm_hidden)
+            self.inf_req_q = ctx.Queue()
+            self.inf_sems = [ctx.Semaphore(0) for _ in range(args.num_actors)]
+            self.weights_version = ctx.Value("l", 0)
+
         # ---- learner state (device init deferred to setup()) ----
         self.learner_model: Optional[AtariNet] = None
         self.global_step = 0
@@ -124,14 +139,40 @@ class ImpalaTrainer:
     def start_actors(self) -> None:
         if self._started:
             return
-        factory = _make_actor_env_factory(self.args)
+        args = self.args
+        factory = _make_actor_env_factory(args)
         ctx = mp.get_context("fork")
-        for i in range(self.args.num_actors):
+
+        if self.inference == "gpu":
+            slots, req_q, sems = self.inf_slots, self.inf_req_q, self.inf_sems
+
+            def make_policy(aid: int):
+                return RemotePolicy(aid, slots, req_q, sems[aid])
+
+            model_kwargs = dict(observation_shape=self.obs_shape,
+                                num_actions=self.num_actions,
+                                use_lstm=args.use_lstm)
+            dev_index = self.device.index or 0
+            self.inference_proc = ctx.Process(
+                target=inference_worker,
+                args=(dev_index, model_kwargs, slots, self.shared_flat.flat,
+                      self.weights_version, req_q, sems, self.stop_event),
+                kwargs=dict(seed=args.seed + 9999),
+                daemon=True, name=f"impala-infer-{self.rank}")
+            self.inference_proc.start()
+        else:
+            actor_model = self.actor_model
+            E = args.envs_per_actor
+
+            def make_policy(aid: int):
+                return LocalPolicy(actor_model, E)
+
+        for i in range(args.num_actors):
             p = ctx.Process(
                 target=actor_loop,
-                args=(i, factory, self.actor_model, self.store, self.free_q,
+                args=(i, factory, make_policy, self.store, self.free_q,
                       self.full_q, self.stop_event, self.step_counter,
-                      self.episode_q, self.args.seed),
+                      self.episode_q, args.seed),
                 daemon=True, name=f"impala-actor-{self.rank}-{i}")
             p.start()
             self.actors.append(p)
@@ -153,21 +194,32 @@ class ImpalaTrainer:
         self.gatherer = BatchGatherer(self.store, self.device,
                                       self.slots_per_batch)
         if self.device.type == "cuda":
+            from ..parallel.rollout import pin_tensor
             pinned = pin_store(self.store)
-            self.log.info(f"pinned {pinned/1e6:.1f} MB of rollout store")
+            pinned += pin_tensor(self.shared_flat.flat)
+            self.log.info(f"pinned {pinned/1e6:.1f} MB (rollout store + "
+                          f"shared weights)")
             self.publish_stream = torch.cuda.Stream(device=self.device)
+            self.publish_event = torch.cuda.Event()
         self._publish_weights()
         self.autocast_dtype = (torch.bfloat16 if args.dtype == "bf16" and
                                self.device.type == "cuda" else None)
 
     @torch.no_grad()
     def _publish_weights(self) -> None:
-        """learner flat → shared CPU flat (one memcpy; actors alias it)."""
+        """learner flat → shared CPU flat (one D2H memcpy on the publish
+        stream; CPU actors alias the buffer, the GPU inference worker
+        reloads it when the version counter bumps)."""
         if self.device.type == "cuda":
+            self.publish_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self.publish_stream):
                 self.shared_flat.flat.copy_(self.flat.flat, non_blocking=True)
+            self.publish_event.record(self.publish_stream)
         else:
             self.shared_flat.flat.copy_(self.flat.flat)
+        if self.inference == "gpu":
+            with self.weights_version.get_lock():
+                self.weights_version.value += 1
 
     # -- core step ---------------------------------------------------------
     def next_batch(self) -> Dict[str, torch.Tensor]:
@@ -219,6 +271,9 @@ class ImpalaTrainer:
 
         all_reduce_flat(self.flat.flat_grad, average=True)
         clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
+        if self.device.type == "cuda":
+            # don't overwrite weights while the previous publish D2H reads them
+            torch.cuda.current_stream().wait_event(self.publish_event)
         self.optimizer.step(self.flat.flat_grad)
         self.timings.time("optimize")
         self._publish_weights()
@@ -226,12 +281,12 @@ class ImpalaTrainer:
 
         self.learn_iters += 1
         self.global_step += args.rollout_length * args.batch_size
-        return {"total_loss": float(total.detach()),
-                "pg_loss": float(comps[0]),
-                "baseline_loss": float(comps[1]),
-                "entropy_loss": float(comps[2])}
+        # stats stay on-device: converting forces a host sync, so callers
+        # float() them only when they actually log
+        return {"total_loss": total.detach(), "pg_loss": comps[0],
+                "baseline_loss": comps[1], "entropy_loss": comps[2]}
 
-    def train_iteration(self) -> Dict[str, float]:
+    def train_iteration(self) -> Dict[str, torch.Tensor]:
         self.timings.reset()
         batch = self.next_batch()
         return self.learn_step(batch)
@@ -258,7 +313,7 @@ class ImpalaTrainer:
                     ret_str = (f" ret={sum(rets)/len(rets):.2f}" if rets else "")
                     self.log.info(
                         f"step {self.global_step} SPS {sps:,.0f} "
-                        f"loss {stats['total_loss']:.3f}{ret_str}")
+                        f"loss {float(stats['total_loss']):.3f}{ret_str}")
                 if (self.rank == 0 and not args.disable_checkpoint and
                         now - last_ckpt > args.checkpoint_interval_s):
                     self.save(ckpt_path)
@@ -299,8 +354,20 @@ class ImpalaTrainer:
         self.stop_event.set()
         for _ in self.actors:
             self.free_q.put(None)  # poison pills (impala_atari.py:480)
+        if self.inference == "gpu":
+            for s in self.inf_sems:
+                s.release()  # unblock any actor waiting on a response
+            try:
+                self.inf_req_q.put_nowait(None)
+            except Exception:
+                pass
         for p in self.actors:
             p.join(timeout=2.0)
             if p.is_alive():
                 p.terminate()
         self.actors.clear()
+        if self.inference_proc is not None:
+            self.inference_proc.join(timeout=2.0)
+            if self.inference_proc.is_alive():
+                self.inference_proc.terminate()
+            self.inference_proc = None

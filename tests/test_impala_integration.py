@@ -24,7 +24,7 @@ def test_impala_cpu_end_to_end(tmp_path):
     try:
         t.start_actors()
         t.setup_learner()
-        losses = [t.train_iteration()["total_loss"] for _ in range(4)]
+        losses = [float(t.train_iteration()["total_loss"]) for _ in range(4)]
         assert all(torch.isfinite(torch.tensor(losses)))
         assert t.global_step == 4 * 8 * 8
         # checkpoint round trip (IMPALA model.tar format)
@@ -59,7 +59,7 @@ def test_impala_no_lstm_path(tmp_path):
         t.start_actors()
         t.setup_learner()
         stats = t.train_iteration()
-        assert torch.isfinite(torch.tensor(stats["total_loss"]))
+        assert torch.isfinite(torch.tensor(float(stats["total_loss"])))
     finally:
         t.shutdown()
 
@@ -82,5 +82,33 @@ def test_impala_learns_synthetic_reward(tmp_path):
                 rews.append(t.next_batch()["reward"].mean().item())
         avg = sum(rews) / len(rews)
         assert avg > 0.22, f"no learning signal: avg reward {avg:.3f}"
+    finally:
+        t.shutdown()
+
+
+@pytest.mark.gpu
+def test_impala_gpu_learner_end_to_end(tmp_path):
+    """GPU learner + CPU actor inference: fused kernels in the loop."""
+    t = ImpalaTrainer(_args(tmp_path, device="cuda:0", dtype="bf16"))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        losses = [float(t.train_iteration()["total_loss"]) for _ in range(4)]
+        assert all(torch.isfinite(torch.tensor(losses)))
+    finally:
+        t.shutdown()
+
+
+@pytest.mark.gpu
+def test_impala_gpu_inference_end_to_end(tmp_path):
+    """SEED-style GPU inference worker feeding the GPU learner."""
+    t = ImpalaTrainer(_args(tmp_path, device="cuda:0", dtype="bf16",
+                            inference="gpu"))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        losses = [float(t.train_iteration()["total_loss"]) for _ in range(4)]
+        assert all(torch.isfinite(torch.tensor(losses)))
+        assert t.inference == "gpu" and t.inference_proc.is_alive()
     finally:
         t.shutdown()
