@@ -207,19 +207,46 @@ def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
     return T * E
 
 
-def actor_loop(actor_id: int, make_env_fn, make_policy_fn, store: RolloutStore,
+def build_actor_env(env_spec: dict, actor_id: int):
+    """Construct an actor's vectorized env from a picklable spec
+    (module-level so spawned processes can build it)."""
+    env_id = env_spec["env_id"]
+    E = env_spec["envs_per_actor"]
+    seed = env_spec.get("seed", 0)
+    if env_id == "synthetic-atari":
+        from ..envs.synthetic import SyntheticAtariVecEnv
+        return SyntheticAtariVecEnv(E, seed=seed * 1000 + actor_id)
+    from ..envs.registry import make_env
+    from ..envs.vec_env import SyncVectorEnv
+    return SyncVectorEnv([
+        (lambda i=i: make_env(env_id, seed=seed * 1000 + actor_id * E + i,
+                              deepmind_wrap=env_spec.get("deepmind_wrap", True)))
+        for i in range(E)])
+
+
+def actor_loop(actor_id: int, env_spec: dict, store: RolloutStore,
                free_q, full_q, stop_event, step_counter,
-               episode_queue=None, seed: int = 0, torch_threads: int = 1):
+               episode_queue=None, seed: int = 0, actor_model=None,
+               inf_slots=None, inf_req_q=None, inf_sem=None,
+               torch_threads: int = 1):
     """Actor process main (reference: impala_atari.py:153-220).
 
-    ``make_policy_fn(actor_id)`` returns either a LocalPolicy (CPU shared
-    model) or a RemotePolicy (GPU inference worker client).  Poison pill: a
-    None on free_q exits.
+    Runs under the *spawn* start method (the learner initializes HIP, so
+    forked children would inherit a poisoned runtime): every argument is
+    picklable; shared tensors travel as shm handles.  Policy: RemotePolicy
+    when inference-slot plumbing is given, else LocalPolicy on
+    ``actor_model`` (whose params alias the published shared flat buffer).
+    Poison pill: a None on free_q exits.
     """
     torch.manual_seed(seed + actor_id)
     torch.set_num_threads(torch_threads)
-    env = make_env_fn(actor_id)
-    policy = make_policy_fn(actor_id)
+    env = build_actor_env(env_spec, actor_id)
+    if inf_slots is not None:
+        from .inference import RemotePolicy
+        policy = RemotePolicy(actor_id, inf_slots, inf_req_q, inf_sem)
+    else:
+        actor_model.eval()
+        policy = LocalPolicy(actor_model, store.envs_per_slot)
     state = ActorState(env, policy, store.envs_per_slot)
     try:
         while not stop_event.is_set():

@@ -35,30 +35,11 @@ from ..envs.synthetic import SyntheticAtariVecEnv
 from ..models.atari import AtariNet
 from ..ops import FusedRMSprop, clip_grad_norm_, impala_loss
 from ..parallel import FlatParams, all_reduce_flat, get_rank, get_world_size
-from ..parallel.inference import (InferenceSlots, RemotePolicy,
-                                  inference_worker)
-from ..parallel.rollout import (ActorState, BatchGatherer, LocalPolicy,
-                                RolloutStore, actor_loop, pin_store)
+from ..parallel.inference import InferenceSlots, inference_worker
+from ..parallel.rollout import (BatchGatherer, RolloutStore, actor_loop,
+                                pin_store)
 from ..utils import Timings, get_logger
 from ..utils.checkpoint import load_checkpoint, save_checkpoint
-
-
-def _make_actor_env_factory(args: ImpalaArguments):
-    env_id = args.env_id
-    E = args.envs_per_actor
-
-    def factory(actor_id: int):
-        if env_id == "synthetic-atari":
-            return SyntheticAtariVecEnv(E, seed=args.seed * 1000 + actor_id)
-        # real envs: vectorize single envs with the DeepMind stack
-        from ..envs.vec_env import SyncVectorEnv
-        from ..envs.registry import make_env
-
-        def one(i):
-            return make_env(env_id, seed=args.seed * 1000 + actor_id * E + i,
-                            deepmind_wrap=True)
-        return SyncVectorEnv([lambda i=i: one(i) for i in range(E)])
-    return factory
 
 
 class ImpalaTrainer:
@@ -106,7 +87,14 @@ class ImpalaTrainer:
             num_buffers, args.rollout_length, E, self.obs_shape,
             self.num_actions, lstm_layers=2, lstm_hidden=lstm_hidden)
 
-        ctx = mp.get_context("fork")
+        # spawn, not fork, whenever the GPU is involved: the learner (and
+        # inference worker) initialize the HIP runtime, which does not
+        # survive fork on ROCm.  Shared-memory tensors travel to spawned
+        # children as shm handles.  Pure-CPU runs keep fork (fast startup).
+        self._mp_ctx = ("fork" if (self.device.type == "cpu"
+                                   and not torch.cuda.is_initialized())
+                        else "spawn")
+        ctx = mp.get_context(self._mp_ctx)
         self.free_q = ctx.SimpleQueue()
         self.full_q = ctx.SimpleQueue()
         self.stop_event = ctx.Event()
@@ -140,40 +128,38 @@ class ImpalaTrainer:
         if self._started:
             return
         args = self.args
-        factory = _make_actor_env_factory(args)
-        ctx = mp.get_context("fork")
+        env_spec = {"env_id": args.env_id,
+                    "envs_per_actor": args.envs_per_actor,
+                    "seed": args.seed + 7919 * self.rank}
+        ctx = mp.get_context(self._mp_ctx)
 
         if self.inference == "gpu":
-            slots, req_q, sems = self.inf_slots, self.inf_req_q, self.inf_sems
-
-            def make_policy(aid: int):
-                return RemotePolicy(aid, slots, req_q, sems[aid])
-
             model_kwargs = dict(observation_shape=self.obs_shape,
                                 num_actions=self.num_actions,
                                 use_lstm=args.use_lstm)
             dev_index = self.device.index or 0
             self.inference_proc = ctx.Process(
                 target=inference_worker,
-                args=(dev_index, model_kwargs, slots, self.shared_flat.flat,
-                      self.weights_version, req_q, sems, self.stop_event),
+                args=(dev_index, model_kwargs, self.inf_slots,
+                      self.shared_flat.flat, self.weights_version,
+                      self.inf_req_q, self.inf_sems, self.stop_event),
                 kwargs=dict(seed=args.seed + 9999),
                 daemon=True, name=f"impala-infer-{self.rank}")
             self.inference_proc.start()
-        else:
-            actor_model = self.actor_model
-            E = args.envs_per_actor
-
-            def make_policy(aid: int):
-                return LocalPolicy(actor_model, E)
 
         for i in range(args.num_actors):
+            kw = dict(seed=args.seed, episode_queue=self.episode_q)
+            if self.inference == "gpu":
+                kw.update(inf_slots=self.inf_slots, inf_req_q=self.inf_req_q,
+                          inf_sem=self.inf_sems[i])
+            else:
+                kw.update(actor_model=self.actor_model)
             p = ctx.Process(
                 target=actor_loop,
-                args=(i, factory, make_policy, self.store, self.free_q,
-                      self.full_q, self.stop_event, self.step_counter,
-                      self.episode_q, args.seed),
-                daemon=True, name=f"impala-actor-{self.rank}-{i}")
+                args=(i, env_spec, self.store, self.free_q, self.full_q,
+                      self.stop_event, self.step_counter),
+                kwargs=kw, daemon=True,
+                name=f"impala-actor-{self.rank}-{i}")
             p.start()
             self.actors.append(p)
         for s in range(self._num_buffers):
