@@ -97,6 +97,7 @@ def main():
 
     for _ in range(args.warmup):
         trainer.train_iteration()
+    trainer.reset_timings()  # drop first-iteration autotune/compile spikes
 
     barrier_sync()
     t0 = time.perf_counter()

@@ -89,34 +89,51 @@ def _libhip():
         return None
 
 
-def pin_tensor(t: torch.Tensor) -> int:
-    """hipHostRegister one CPU tensor in this process (0 on failure)."""
-    lib = _libhip()
-    if lib is None or not torch.cuda.is_available():
-        return 0
-    nbytes = t.numel() * t.element_size()
-    if lib.hipHostRegister(ctypes.c_void_p(t.data_ptr()),
-                           ctypes.c_size_t(nbytes), ctypes.c_uint(0)) == 0:
-        return nbytes
-    return 0
+class PinRegistry:
+    """Tracks hipHostRegister'd ranges so they can be unregistered BEFORE
+    the underlying shared-memory segments are freed — a stale registration
+    aborts the HIP runtime on a later allocation/copy."""
 
+    def __init__(self):
+        self._ranges = []  # (ptr, nbytes)
 
-def pin_store(store: RolloutStore) -> int:
-    """hipHostRegister every shared slot tensor in THIS process so H2D
-    copies from them are async DMA.  Registration is per-process; call it
-    from the learner after CUDA init.  Returns bytes pinned (0 = no-op)."""
-    lib = _libhip()
-    if lib is None or not torch.cuda.is_available():
-        return 0
-    pinned = 0
-    for t in store.fields().values():
-        ptr = t.data_ptr()
+    def pin(self, t: torch.Tensor) -> int:
+        lib = _libhip()
+        if lib is None or not torch.cuda.is_available():
+            return 0
         nbytes = t.numel() * t.element_size()
+        ptr = t.data_ptr()
         # hipHostRegisterDefault = 0
         if lib.hipHostRegister(ctypes.c_void_p(ptr), ctypes.c_size_t(nbytes),
                                ctypes.c_uint(0)) == 0:
-            pinned += nbytes
-    return pinned
+            self._ranges.append((ptr, nbytes))
+            return nbytes
+        return 0
+
+    def pin_store(self, store: "RolloutStore") -> int:
+        return sum(self.pin(t) for t in store.fields().values())
+
+    def unpin_all(self) -> None:
+        lib = _libhip()
+        if lib is None:
+            return
+        for ptr, _ in self._ranges:
+            lib.hipHostUnregister(ctypes.c_void_p(ptr))
+        self._ranges.clear()
+
+
+def pin_tensor(t: torch.Tensor, registry: "PinRegistry" = None) -> int:
+    """hipHostRegister one CPU tensor in this process (0 on failure)."""
+    reg = registry or PinRegistry()
+    return reg.pin(t)
+
+
+def pin_store(store: RolloutStore, registry: "PinRegistry" = None) -> int:
+    """hipHostRegister every shared slot tensor in THIS process so H2D
+    copies from them are async DMA.  Registration is per-process; call it
+    from the learner after CUDA init.  Returns bytes pinned (0 = no-op)."""
+    reg = registry or PinRegistry()
+    return reg.pin_store(store)
 
 
 class LocalPolicy:

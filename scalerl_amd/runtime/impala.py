@@ -180,9 +180,12 @@ class ImpalaTrainer:
         self.gatherer = BatchGatherer(self.store, self.device,
                                       self.slots_per_batch)
         if self.device.type == "cuda":
-            from ..parallel.rollout import pin_tensor
-            pinned = pin_store(self.store)
-            pinned += pin_tensor(self.shared_flat.flat)
+            # MIOpen exhaustive-find once per conv shape, then cached
+            torch.backends.cudnn.benchmark = True
+            from ..parallel.rollout import PinRegistry
+            self._pins = PinRegistry()
+            pinned = self._pins.pin_store(self.store)
+            pinned += self._pins.pin(self.shared_flat.flat)
             self.log.info(f"pinned {pinned/1e6:.1f} MB (rollout store + "
                           f"shared weights)")
             self.publish_stream = torch.cuda.Stream(device=self.device)
@@ -277,6 +280,9 @@ class ImpalaTrainer:
         batch = self.next_batch()
         return self.learn_step(batch)
 
+    def reset_timings(self) -> None:
+        self.timings = Timings()
+
     # -- driver ------------------------------------------------------------
     def train(self) -> None:
         args = self.args
@@ -357,3 +363,8 @@ class ImpalaTrainer:
             if self.inference_proc.is_alive():
                 self.inference_proc.terminate()
             self.inference_proc = None
+        if getattr(self, "_pins", None) is not None:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            self._pins.unpin_all()
+            self._pins = None
