@@ -290,7 +290,16 @@ def actor_loop(actor_id: int, env_spec: dict, store: RolloutStore,
 class BatchGatherer:
     """Learner-side: gather K slots into device batch tensors [T+1, K*E, ...]
     via side-stream async H2D copies (the device boundary of the pipeline,
-    replacing impala_atari.py:248-266)."""
+    replacing impala_atari.py:248-266).
+
+    Double-buffered: ``start(slot_ids)`` launches the copies into the next
+    staging set and returns a token; ``finish(token)`` host-syncs the copy
+    event (slots are then recyclable) and returns the batch views.  The
+    trainer starts batch N+1's copies before computing batch N, so H2D
+    rides the side stream under the learner's kernels.
+    """
+
+    NBUF = 2
 
     def __init__(self, store: RolloutStore, device: torch.device,
                  slots_per_batch: int):
@@ -298,42 +307,70 @@ class BatchGatherer:
         self.device = device
         self.K = slots_per_batch
         self.is_cuda = device.type == "cuda"
-        T1 = store.rollout_length + 1
-        E = store.envs_per_slot
         K = self.K
+        self._next_buf = 0
         if self.is_cuda:
             self.stream = torch.cuda.Stream(device=device)
-            self.staging = {
-                name: torch.empty((K, *t.shape[1:]), dtype=t.dtype,
-                                  device=device)
-                for name, t in store.fields().items()
-            }
-            self.ready_event = torch.cuda.Event()
+            self.staging = [
+                {name: torch.empty((K, *t.shape[1:]), dtype=t.dtype,
+                                   device=device)
+                 for name, t in store.fields().items()}
+                for _ in range(self.NBUF)
+            ]
+            self.events = [torch.cuda.Event() for _ in range(self.NBUF)]
+            # recorded on the MAIN stream when the learner is done reading a
+            # staging set; the copy stream waits on it before overwriting
+            self.consumed = [torch.cuda.Event() for _ in range(self.NBUF)]
+            self._last_buf = None
 
-    def gather(self, slot_ids: List[int]) -> Dict[str, torch.Tensor]:
-        """Copy slots to device; returns batch dict with [T+1, B, ...] layout
-        (B = K*E).  Caller may recycle the slots after `copies_done()`."""
-        store = self.store
+    def start(self, slot_ids: List[int]):
+        """Launch async H2D of the given slots; returns a token."""
         if not self.is_cuda:
-            batch = {}
-            for name, t in store.fields().items():
-                stacked = torch.stack([t[s] for s in slot_ids])  # [K, ...]
-                batch[name] = self._to_batch_layout(name, stacked)
-            return batch
+            return ("cpu", list(slot_ids))
+        buf = self._next_buf
+        self._next_buf = (buf + 1) % self.NBUF
         with torch.cuda.stream(self.stream):
-            for name, t in store.fields().items():
-                dst = self.staging[name]
+            # don't overwrite a staging set the learner might still read
+            self.stream.wait_event(self.consumed[buf])
+            staging = self.staging[buf]
+            for name, t in self.store.fields().items():
+                dst = staging[name]
                 for j, s in enumerate(slot_ids):
                     dst[j].copy_(t[s], non_blocking=True)
-            self.ready_event.record(self.stream)
-        torch.cuda.current_stream(self.device).wait_event(self.ready_event)
-        return {name: self._to_batch_layout(name, self.staging[name])
-                for name in self.staging}
+            self.events[buf].record(self.stream)
+        return ("cuda", buf)
+
+    def finish(self, token) -> Dict[str, torch.Tensor]:
+        """Host-sync the token's copies (slots reusable afterwards) and
+        return the batch in [T+1, B, ...] layout."""
+        kind, payload = token
+        if kind == "cpu":
+            batch = {}
+            for name, t in self.store.fields().items():
+                stacked = torch.stack([t[s] for s in payload])
+                batch[name] = self._to_batch_layout(name, stacked)
+            return batch
+        buf = payload
+        self.events[buf].synchronize()
+        torch.cuda.current_stream(self.device).wait_event(self.events[buf])
+        self._last_buf = buf
+        return {name: self._to_batch_layout(name, self.staging[buf][name])
+                for name in self.staging[buf]}
+
+    def mark_consumed(self) -> None:
+        """Record (on the main stream) that all work reading the last
+        finished staging set has been enqueued; called by the learner after
+        its step so the next prefetch into that set can proceed."""
+        if self.is_cuda and self._last_buf is not None:
+            self.consumed[self._last_buf].record(
+                torch.cuda.current_stream(self.device))
+
+    def gather(self, slot_ids: List[int]) -> Dict[str, torch.Tensor]:
+        """Synchronous convenience path (tests, non-pipelined callers)."""
+        return self.finish(self.start(slot_ids))
 
     def copies_done(self) -> None:
-        """Block host until the H2D copies finished (slots reusable)."""
-        if self.is_cuda:
-            self.ready_event.synchronize()
+        """Back-compat no-op: finish() already syncs the copy event."""
 
     def _to_batch_layout(self, name: str, stacked: torch.Tensor) -> torch.Tensor:
         """[K, T+1, E, ...] → [T+1, K*E, ...]; core_state → [L, K*E, H] x2."""

@@ -122,6 +122,7 @@ class ImpalaTrainer:
         self.global_step = 0
         self.learn_iters = 0
         self.timings = Timings()
+        self._pending = None  # in-flight prefetched batch token
 
     # -- lifecycle ---------------------------------------------------------
     def start_actors(self) -> None:
@@ -211,15 +212,24 @@ class ImpalaTrainer:
                 self.weights_version.value += 1
 
     # -- core step ---------------------------------------------------------
-    def next_batch(self) -> Dict[str, torch.Tensor]:
+    def _start_prefetch(self):
         slot_ids = [self.full_q.get() for _ in range(self.slots_per_batch)]
         self.timings.time("dequeue")
-        batch = self.gatherer.gather(slot_ids)
-        self.timings.time("gather")
-        self.gatherer.copies_done()
+        token = self.gatherer.start(slot_ids)
+        self.timings.time("gather_start")
+        return (token, slot_ids)
+
+    def next_batch(self) -> Dict[str, torch.Tensor]:
+        """Finish the pending prefetch (or do a cold gather) and kick off
+        the next one — batch N+1's H2D copies overlap batch N's compute."""
+        if self._pending is None:
+            self._pending = self._start_prefetch()
+        token, slot_ids = self._pending
+        batch = self.gatherer.finish(token)
+        self.timings.time("gather_wait")
         for s in slot_ids:
             self.free_q.put(s)
-        self.timings.time("enqueue")
+        self._pending = self._start_prefetch()
         return batch
 
     def learn_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
@@ -268,6 +278,7 @@ class ImpalaTrainer:
         self._publish_weights()
         self.timings.time("publish")
 
+        self.gatherer.mark_consumed()
         self.learn_iters += 1
         self.global_step += args.rollout_length * args.batch_size
         # stats stay on-device: converting forces a host sync, so callers

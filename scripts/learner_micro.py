@@ -7,9 +7,13 @@ full actor-learner pipeline); it measures the learner's ceiling.
 """
 
 import argparse
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from scalerl_amd.models import AtariNet
 from scalerl_amd.ops import FusedRMSprop, clip_grad_norm_, impala_loss
