@@ -158,6 +158,7 @@ class ImpalaArguments(RLArguments):
     checkpoint_path: str = _h("explicit checkpoint file (empty → auto)", default="")
     inference: str = _h("actor inference placement: cpu | gpu", default="cpu")
     dtype: str = _h("learner compute dtype: bf16 | fp32", default="bf16")
+    use_graph: bool = _h("hipGraph-capture the learner step", default=True)
 
 
 @dataclass

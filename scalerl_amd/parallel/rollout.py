@@ -299,7 +299,8 @@ class BatchGatherer:
     rides the side stream under the learner's kernels.
     """
 
-    NBUF = 2
+    NBUF = 3  # 3 staging sets: copy into buf i+1 never fences on the
+    # learner step that is currently executing (2 would serialize them)
 
     def __init__(self, store: RolloutStore, device: torch.device,
                  slots_per_batch: int):

@@ -1,0 +1,84 @@
+"""hipGraph-captured learner step.
+
+The IMPALA learner's forward+backward is hundreds of small kernel launches
+(the MaskedLSTM unrolls 2 layers × (T+1) steps, each a GEMM + a fused
+pointwise kernel) — measured on MI355X the HOST enqueue time dominates the
+iteration (~40 ms of Python/launch per step at B=128).  Capturing
+{zero-grad → model fwd → fused V-trace/loss → backward} into one hipGraph
+replays the whole step as a single launch; the fused HIP kernels
+(ctypes-launched onto the capture stream) capture along with torch's.
+
+Kept OUTSIDE the graph: RCCL all-reduce (watchdog interaction), grad clip,
+fused optimizer, weight publish — five launches, microseconds of enqueue.
+
+Inputs are copied D2D from the gatherer's staging set into this object's
+static buffers before each replay (cheap: device bandwidth).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Tuple
+
+import torch
+
+
+class GraphedImpalaStep:
+    def __init__(self, model, flat_grad: torch.Tensor, loss_kwargs: Dict,
+                 batch_example: Dict[str, torch.Tensor], use_lstm: bool,
+                 autocast_dtype, reward_clip: bool, discounting: float,
+                 warmup_iters: int = 3):
+        from ..ops import impala_loss
+
+        self.model = model
+        self.flat_grad = flat_grad
+        self.use_lstm = use_lstm
+        dev = flat_grad.device
+
+        # static input buffers (graph reads these addresses every replay)
+        self.static: Dict[str, torch.Tensor] = {
+            k: torch.empty_like(v) for k, v in batch_example.items()
+        }
+
+        def run_step():
+            self.flat_grad.zero_()
+            inputs = {k: self.static[k]
+                      for k in ("obs", "reward", "done", "last_action")}
+            core_state = ()
+            if use_lstm:
+                cs = self.static["core_state"]
+                core_state = (cs[0], cs[1])
+            if autocast_dtype is not None:
+                with torch.autocast(device_type="cuda", dtype=autocast_dtype):
+                    out, _ = model(inputs, core_state, greedy=True)
+            else:
+                out, _ = model(inputs, core_state, greedy=True)
+            logits = out["policy_logits"].float()
+            baseline = out["baseline"].float()
+            rewards = self.static["reward"][1:]
+            if reward_clip:
+                rewards = torch.clamp(rewards, -1, 1)
+            discounts = (~self.static["done"][1:]).float() * discounting
+            total, comps, _ = impala_loss(
+                self.static["logits"][:-1], logits[:-1],
+                self.static["action"][:-1], rewards, discounts,
+                baseline[:-1], baseline[-1].detach(), **loss_kwargs)
+            total.backward()
+            return total.detach(), comps
+
+        # torch-required warmup on a side stream before capture
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(s):
+            for _ in range(warmup_iters):
+                run_step()
+        torch.cuda.current_stream(dev).wait_stream(s)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
+            self.total, self.comps = run_step()
+
+    def run(self, batch: Dict[str, torch.Tensor]) -> Tuple[torch.Tensor, torch.Tensor]:
+        for k, dst in self.static.items():
+            dst.copy_(batch[k], non_blocking=True)
+        self.graph.replay()
+        return self.total, self.comps

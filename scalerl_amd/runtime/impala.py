@@ -123,6 +123,8 @@ class ImpalaTrainer:
         self.learn_iters = 0
         self.timings = Timings()
         self._pending = None  # in-flight prefetched batch token
+        self._graphed = None
+        self.use_graph = bool(args.use_graph) and self.device.type == "cuda"
 
     # -- lifecycle ---------------------------------------------------------
     def start_actors(self) -> None:
@@ -232,12 +234,18 @@ class ImpalaTrainer:
         self._pending = self._start_prefetch()
         return batch
 
-    def learn_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+    def _loss_kwargs(self) -> Dict[str, float]:
+        args = self.args
+        return dict(clip_rho_threshold=args.clip_rho_threshold,
+                    clip_c_threshold=1.0,
+                    clip_pg_rho_threshold=args.clip_pg_rho_threshold,
+                    baseline_cost=args.baseline_cost,
+                    entropy_cost=args.entropy_cost)
+
+    def _eager_fwd_bwd(self, batch):
         args = self.args
         model = self.learner_model
-        model.train()
         self.flat.flat_grad.zero_()
-
         inputs = {"obs": batch["obs"], "reward": batch["reward"],
                   "done": batch["done"], "last_action": batch["last_action"]}
         core_state = ()
@@ -251,22 +259,35 @@ class ImpalaTrainer:
             out, _ = model(inputs, core_state, greedy=True)
         logits = out["policy_logits"].float()
         baseline = out["baseline"].float()
-
         rewards = batch["reward"][1:]
         if args.reward_clipping == "abs_one":
             rewards = torch.clamp(rewards, -1, 1)
         discounts = (~batch["done"][1:]).float() * args.discounting
-
         total, comps, _ = impala_loss(
             batch["logits"][:-1], logits[:-1], batch["action"][:-1],
             rewards, discounts, baseline[:-1], baseline[-1].detach(),
-            clip_rho_threshold=args.clip_rho_threshold,
-            clip_c_threshold=1.0,
-            clip_pg_rho_threshold=args.clip_pg_rho_threshold,
-            baseline_cost=args.baseline_cost, entropy_cost=args.entropy_cost)
+            **self._loss_kwargs())
         self.timings.time("forward")
         total.backward()
         self.timings.time("backward")
+        return total.detach(), comps
+
+    def learn_step(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        args = self.args
+        self.learner_model.train()
+        if self.use_graph:
+            if self._graphed is None:
+                from .graphed import GraphedImpalaStep
+                self.log.info("capturing learner step into a hipGraph …")
+                self._graphed = GraphedImpalaStep(
+                    self.learner_model, self.flat.flat_grad,
+                    self._loss_kwargs(), batch, args.use_lstm,
+                    self.autocast_dtype,
+                    args.reward_clipping == "abs_one", args.discounting)
+            total, comps = self._graphed.run(batch)
+            self.timings.time("graph_replay")
+        else:
+            total, comps = self._eager_fwd_bwd(batch)
 
         all_reduce_flat(self.flat.flat_grad, average=True)
         clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
@@ -344,9 +365,10 @@ class ImpalaTrainer:
 
     def load(self, path: str) -> None:
         ckpt = load_checkpoint(path, map_location=self.device)
+        # load_state_dict copies in-place into the existing params, which
+        # are views of the flat buffer — the flat pair (and any captured
+        # graph / optimizer state bound to it) stays valid
         self.learner_model.load_state_dict(ckpt["model_state_dict"])
-        # load_state_dict replaced param storages; re-flatten
-        self.flat = FlatParams(self.learner_model, device=self.device)
         opt_sd = ckpt.get("optimizer_state_dict")
         if opt_sd:
             self.optimizer.load_state_dict(opt_sd)
