@@ -34,9 +34,11 @@ class GraphedImpalaStep:
         self.use_lstm = use_lstm
         dev = flat_grad.device
 
-        # static input buffers (graph reads these addresses every replay)
+        # static input buffers (graph reads these addresses every replay),
+        # initialized from the example batch — uninitialized int fields
+        # (actions) would index out of bounds during warmup
         self.static: Dict[str, torch.Tensor] = {
-            k: torch.empty_like(v) for k, v in batch_example.items()
+            k: v.clone() for k, v in batch_example.items()
         }
 
         def run_step():
