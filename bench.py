@@ -30,8 +30,10 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--rollout-length", type=int, default=80)
-    p.add_argument("--batch-size", type=int, default=32)
-    p.add_argument("--envs-per-actor", type=int, default=16)
+    p.add_argument("--use-graph", type=int, default=0,
+                   help="hipGraph-capture the learner step (experimental)")
+    p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--envs-per-actor", type=int, default=128)
     p.add_argument("--num-actors", type=int, default=0,
                    help="actor procs per rank (0 = auto from cpu count)")
     p.add_argument("--use-lstm", type=int, default=1)
@@ -65,7 +67,7 @@ def main():
         if inference == "gpu":
             # actors are pure env-steppers; a handful saturate the
             # inference worker
-            args.num_actors = min(16, avail)
+            args.num_actors = min(24, avail)
         else:
             # one CPU-inference actor ≈ a few hundred steps/s
             args.num_actors = min(64, avail)
@@ -74,6 +76,7 @@ def main():
         rollout_length=args.rollout_length, batch_size=args.batch_size,
         envs_per_actor=args.envs_per_actor, num_actors=args.num_actors,
         use_lstm=bool(args.use_lstm), device=device, dtype=args.dtype,
+        use_graph=bool(args.use_graph),
         inference=inference, seed=1234 + rank,
         total_steps=1 << 60, disable_checkpoint=True,
         output_dir="/tmp/scalerl_bench")

@@ -158,7 +158,9 @@ class ImpalaArguments(RLArguments):
     checkpoint_path: str = _h("explicit checkpoint file (empty → auto)", default="")
     inference: str = _h("actor inference placement: cpu | gpu", default="cpu")
     dtype: str = _h("learner compute dtype: bf16 | fp32", default="bf16")
-    use_graph: bool = _h("hipGraph-capture the learner step", default=True)
+    use_graph: bool = _h("hipGraph-capture the learner step "
+                         "(experimental: conflicts with the GPU inference "
+                         "worker on ROCm 7.2)", default=False)
 
 
 @dataclass
