@@ -46,6 +46,9 @@ def parse_args():
 
 
 def main():
+    # Consistent MIOpen behavior across boxes: full find (the hybrid default
+    # has been observed to settle on naive bf16 conv kernels on fresh boxes).
+    os.environ.setdefault("MIOPEN_FIND_MODE", "1")
     args = parse_args()
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

@@ -154,42 +154,41 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
                 seen_version = version.value
                 flat.load_from(shared_flat, non_blocking=False)
 
-            k = len(ids)
-            # H2D for the requesting actors
-            for j, a in enumerate(ids):
-                obs_d[j].copy_(slots.obs[a], non_blocking=True)
-                rew_d[j].copy_(slots.reward[a], non_blocking=True)
-                done_d[j].copy_(slots.done[a], non_blocking=True)
-                lastact_d[j].copy_(slots.last_action[a], non_blocking=True)
-            cols = torch.tensor([a * E + e for a in ids for e in range(E)],
-                                device=device)
+            # FIXED-SHAPE round: forward ALL A actor slots every time (only
+            # requesting actors' inputs are refreshed and only their states/
+            # outputs are written back).  A varying batch size would retrigger
+            # MIOpen find per new shape and preclude graph capture.
+            for a in ids:
+                obs_d[a].copy_(slots.obs[a], non_blocking=True)
+                rew_d[a].copy_(slots.reward[a], non_blocking=True)
+                done_d[a].copy_(slots.done[a], non_blocking=True)
+                lastact_d[a].copy_(slots.last_action[a], non_blocking=True)
             inputs = {
-                "obs": obs_d[:k].reshape(1, k * E, *slots.obs.shape[2:]),
-                "reward": rew_d[:k].reshape(1, k * E),
-                "done": done_d[:k].reshape(1, k * E),
-                "last_action": lastact_d[:k].reshape(1, k * E),
+                "obs": obs_d.reshape(1, A * E, *slots.obs.shape[2:]),
+                "reward": rew_d.reshape(1, A * E),
+                "done": done_d.reshape(1, A * E),
+                "last_action": lastact_d.reshape(1, A * E),
             }
             state = ()
             snap = {}
             if use_lstm:
-                h = h_all.index_select(1, cols)
-                c = c_all.index_select(1, cols)
-                for j, a in enumerate(ids):
+                for a in ids:
                     if slots.want_state[a]:
-                        snap[a] = (j, h[:, j * E:(j + 1) * E].clone(),
-                                   c[:, j * E:(j + 1) * E].clone())
-                state = (h, c)
+                        snap[a] = (h_all[:, a * E:(a + 1) * E].clone(),
+                                   c_all[:, a * E:(a + 1) * E].clone())
+                state = (h_all.clone(), c_all.clone())
             out, new_state = model(inputs, state)
-            if use_lstm:
-                h_all.index_copy_(1, cols, new_state[0])
-                c_all.index_copy_(1, cols, new_state[1])
-            action = out["action"].view(k, E)
-            logits = out["policy_logits"].view(k, E, nact)
-            for j, a in enumerate(ids):
-                slots.action[a].copy_(action[j], non_blocking=True)
-                slots.logits[a].copy_(logits[j], non_blocking=True)
+            action = out["action"].view(A, E)
+            logits = out["policy_logits"].view(A, E, nact)
+            for a in ids:
+                if use_lstm:
+                    sl = slice(a * E, (a + 1) * E)
+                    h_all[:, sl] = new_state[0][:, sl]
+                    c_all[:, sl] = new_state[1][:, sl]
+                slots.action[a].copy_(action[a], non_blocking=True)
+                slots.logits[a].copy_(logits[a], non_blocking=True)
                 if a in snap:
-                    _, hj, cj = snap[a]
+                    hj, cj = snap[a]
                     slots.core_state[a][0].copy_(hj, non_blocking=True)
                     slots.core_state[a][1].copy_(cj, non_blocking=True)
             torch.cuda.synchronize()

@@ -21,6 +21,7 @@ from scalerl_amd.parallel import FlatParams
 
 
 def main():
+    os.environ.setdefault("MIOPEN_FIND_MODE", "1")
     p = argparse.ArgumentParser()
     p.add_argument("--rollout-length", type=int, default=80)
     p.add_argument("--batch-size", type=int, default=32)
