@@ -1,7 +1,7 @@
-from .atari import AtariNet
+from .atari import AtariNet, AtariQNet
 from .mlp import ActorCriticNet, ActorNet, CriticNet, QNet
 from .a3c_atari import A3CAtariNet
 from .resnet import ResNetLSTMPolicy
 
-__all__ = ["AtariNet", "QNet", "ActorNet", "CriticNet", "ActorCriticNet",
+__all__ = ["AtariNet", "AtariQNet", "QNet", "ActorNet", "CriticNet", "ActorCriticNet",
            "A3CAtariNet", "ResNetLSTMPolicy"]

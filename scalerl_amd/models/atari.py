@@ -102,3 +102,36 @@ class AtariNet(nn.Module):
         return ({"policy_logits": policy_logits.view(T, B, self.num_actions),
                  "baseline": baseline.view(T, B),
                  "action": action.view(T, B)}, core_state)
+
+
+class AtariQNet(nn.Module):
+    """Nature-CNN Q-network for Ape-X (apex/network.py parity, dueling
+    option): uint8 [N,C,84,84] → Q [N,A]."""
+
+    def __init__(self, observation_shape=(4, 84, 84), num_actions: int = 6,
+                 dueling: bool = True):
+        super().__init__()
+        c = observation_shape[0]
+        self.num_actions = num_actions
+        self.dueling = dueling
+        self.conv1 = nn.Conv2d(c, 32, kernel_size=8, stride=4)
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
+        self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
+        self.fc = nn.Linear(64 * 7 * 7, 512)
+        if dueling:
+            self.value_head = nn.Linear(512, 1)
+            self.adv_head = nn.Linear(512, num_actions)
+        else:
+            self.head = nn.Linear(512, num_actions)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = x.float() / 255.0
+        x = F.relu(self.conv1(x))
+        x = F.relu(self.conv2(x))
+        x = F.relu(self.conv3(x))
+        h = F.relu(self.fc(torch.flatten(x, 1)))
+        if self.dueling:
+            v = self.value_head(h)
+            a = self.adv_head(h)
+            return v + a - a.mean(dim=-1, keepdim=True)
+        return self.head(h)
