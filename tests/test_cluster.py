@@ -1,0 +1,62 @@
+"""TCP control plane over localhost: handshake, weight pull, episode push."""
+
+import threading
+
+import torch
+
+from scalerl_amd.parallel.cluster import (FrameConnection, ParameterServer,
+                                          RemoteWorkerCluster, WorkerServer)
+
+
+def test_parameter_server_versioning():
+    ps = ParameterServer()
+    w, v = ps.pull()
+    assert w is None and v == 0
+    ps.push(torch.ones(5))
+    w, v = ps.pull(have_version=-1)
+    assert v == 1 and w.sum() == 5
+    w2, v2 = ps.pull(have_version=1)  # up to date → no payload
+    assert w2 is None and v2 == 1
+
+
+def test_server_client_roundtrip():
+    srv = WorkerServer({"env_id": "synthetic-atari", "rollout": 8}, port=0)
+    try:
+        srv.publish_weights(torch.arange(10, dtype=torch.float32))
+        c1 = RemoteWorkerCluster("127.0.0.1", srv.port)
+        c2 = RemoteWorkerCluster("127.0.0.1", srv.port)
+        assert {c1.worker_id, c2.worker_id} == {0, 1}
+        assert c1.config["env_id"] == "synthetic-atari"
+
+        w = c1.pull_weights()
+        torch.testing.assert_close(w, torch.arange(10, dtype=torch.float32))
+        # cached: second pull without republish returns same tensor
+        assert c1.pull_weights() is w
+
+        def generate(config, weights):
+            obs = torch.randint(0, 255, (config["rollout"], 4), dtype=torch.uint8)
+            rew = torch.rand(config["rollout"])
+            return {"env_steps": config["rollout"]}, [obs, rew]
+
+        c1.run(generate, iterations=3)
+        c2.run(generate, iterations=2)
+        assert len(srv.episodes) == 5
+        hdr, tensors = srv.episodes[0]
+        assert hdr["env_steps"] == 8
+        assert tensors[0].shape == (8, 4) and tensors[0].dtype == torch.uint8
+        c1.close()
+        c2.close()
+    finally:
+        srv.close()
+
+
+def test_frame_connection_large_tensor():
+    srv = WorkerServer({}, port=0)
+    try:
+        srv.publish_weights(torch.randn(1_000_000))  # 4 MB blob
+        c = RemoteWorkerCluster("127.0.0.1", srv.port)
+        w = c.pull_weights()
+        assert w.numel() == 1_000_000
+        c.close()
+    finally:
+        srv.close()
