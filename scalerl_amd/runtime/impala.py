@@ -347,6 +347,48 @@ class ImpalaTrainer:
                 self.save(ckpt_path)
             self.shutdown()
 
+    @torch.no_grad()
+    def evaluate(self, num_episodes: int = 3,
+                 max_steps: int = 10_000) -> Dict[str, float]:
+        """Greedy-policy evaluation with the current published weights on a
+        fresh single env (reference test-mode parity; runs on the CPU actor
+        model so it never perturbs the learner)."""
+        from ..envs.registry import make_env
+        deepmind = self.args.env_id not in ("synthetic-atari",)
+        env = make_env(self.args.env_id, seed=self.args.seed + 4242,
+                       deepmind_wrap=deepmind)
+        model = self.actor_model
+        model.eval()
+        returns, lengths = [], []
+        for ep in range(num_episodes):
+            obs, _ = env.reset()
+            state = model.initial_state(1)
+            reward, done, last_action = 0.0, True, 0
+            total, steps = 0.0, 0
+            while steps < max_steps:
+                inputs = {
+                    "obs": torch.from_numpy(obs).view(1, 1, *obs.shape),
+                    "reward": torch.tensor([[reward]], dtype=torch.float32),
+                    "done": torch.tensor([[done]]),
+                    "last_action": torch.tensor([[last_action]]),
+                }
+                out, state = model(inputs, state, greedy=True)
+                a = int(out["action"].item())
+                obs, reward, term, trunc, _ = env.step(a)
+                total += reward
+                steps += 1
+                done = term or trunc
+                last_action = a
+                if done:
+                    break
+            returns.append(total)
+            lengths.append(steps)
+        env.close()
+        import numpy as np
+        return {"reward_mean": float(np.mean(returns)),
+                "reward_std": float(np.std(returns)),
+                "length_mean": float(np.mean(lengths))}
+
     def _drain_episode_returns(self) -> List[float]:
         rets: List[float] = []
         try:

@@ -3,9 +3,13 @@ from .schedulers import LinearDecayScheduler, MultiStepScheduler, PiecewiseSched
 from .timings import Timings, Timer
 from .model_utils import hard_target_update, soft_target_update
 from .checkpoint import load_checkpoint, save_checkpoint
+from .progress import ProgressBar, track_parallel_progress, track_progress
+from .algo_utils import chkpt_attribute_to_device, compile_model, remove_compile_prefix
 
 __all__ = [
     "get_logger", "LinearDecayScheduler", "MultiStepScheduler",
     "PiecewiseScheduler", "Timings", "Timer", "hard_target_update",
     "soft_target_update", "save_checkpoint", "load_checkpoint",
+    "ProgressBar", "track_progress", "track_parallel_progress",
+    "chkpt_attribute_to_device", "compile_model", "remove_compile_prefix",
 ]
