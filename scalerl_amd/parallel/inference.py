@@ -27,6 +27,7 @@ from __future__ import annotations
 
 import ctypes
 import queue
+import time
 from typing import Dict, List, Optional, Tuple
 
 import numpy as np
@@ -99,7 +100,7 @@ class RemotePolicy:
 def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlots,
                      shared_flat: torch.Tensor, version: "mp.Value",
                      req_q, sems: List, stop_event, max_batch_actors: int = 0,
-                     seed: int = 0):
+                     seed: int = 0, pause_flag=None, paused_ack=None):
     """Inference worker process main."""
     from ..models.atari import AtariNet
     from .flat import FlatParams
@@ -134,6 +135,19 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
 
     with torch.no_grad():
         while not stop_event.is_set():
+            if pause_flag is not None and pause_flag.value:
+                # learner is hipGraph-capturing on this device: quiesce our
+                # HIP queue entirely until it clears the flag (concurrent
+                # submissions from another process during stream capture
+                # fault the HSA queue on ROCm 7.2)
+                torch.cuda.synchronize()
+                if paused_ack is not None:
+                    paused_ack.value = 1
+                while pause_flag.value and not stop_event.is_set():
+                    time.sleep(0.002)
+                if paused_ack is not None:
+                    paused_ack.value = 0
+                continue
             try:
                 first = req_q.get(timeout=0.2)
             except queue.Empty:

@@ -116,6 +116,8 @@ class ImpalaTrainer:
             self.inf_req_q = ctx.Queue()
             self.inf_sems = [ctx.Semaphore(0) for _ in range(args.num_actors)]
             self.weights_version = ctx.Value("l", 0)
+            self.inf_pause = ctx.Value("i", 0)
+            self.inf_paused_ack = ctx.Value("i", 0)
 
         # ---- learner state (device init deferred to setup()) ----
         self.learner_model: Optional[AtariNet] = None
@@ -146,7 +148,8 @@ class ImpalaTrainer:
                 args=(dev_index, model_kwargs, self.inf_slots,
                       self.shared_flat.flat, self.weights_version,
                       self.inf_req_q, self.inf_sems, self.stop_event),
-                kwargs=dict(seed=args.seed + 9999),
+                kwargs=dict(seed=args.seed + 9999, pause_flag=self.inf_pause,
+                            paused_ack=self.inf_paused_ack),
                 daemon=True, name=f"impala-infer-{self.rank}")
             self.inference_proc.start()
 
@@ -196,6 +199,18 @@ class ImpalaTrainer:
         self._publish_weights()
         self.autocast_dtype = (torch.bfloat16 if args.dtype == "bf16" and
                                self.device.type == "cuda" else None)
+
+    def _pause_inference(self, pause: bool, timeout_s: float = 30.0) -> None:
+        """Quiesce the inference worker's HIP queue around graph capture."""
+        if self.inference != "gpu":
+            return
+        if pause:
+            self.inf_pause.value = 1
+            deadline = time.time() + timeout_s
+            while not self.inf_paused_ack.value and time.time() < deadline:
+                time.sleep(0.005)
+        else:
+            self.inf_pause.value = 0
 
     @torch.no_grad()
     def _publish_weights(self) -> None:
@@ -279,11 +294,15 @@ class ImpalaTrainer:
             if self._graphed is None:
                 from .graphed import GraphedImpalaStep
                 self.log.info("capturing learner step into a hipGraph …")
-                self._graphed = GraphedImpalaStep(
-                    self.learner_model, self.flat.flat_grad,
-                    self._loss_kwargs(), batch, args.use_lstm,
-                    self.autocast_dtype,
-                    args.reward_clipping == "abs_one", args.discounting)
+                self._pause_inference(True)
+                try:
+                    self._graphed = GraphedImpalaStep(
+                        self.learner_model, self.flat.flat_grad,
+                        self._loss_kwargs(), batch, args.use_lstm,
+                        self.autocast_dtype,
+                        args.reward_clipping == "abs_one", args.discounting)
+                finally:
+                    self._pause_inference(False)
             total, comps = self._graphed.run(batch)
             self.timings.time("graph_replay")
         else:
