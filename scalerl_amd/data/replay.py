@@ -215,8 +215,10 @@ class PrioritizedReplayBuffer(ReplayBuffer):
         self.max_priority = 1.0
 
     def add_batch(self, obs, action, reward, next_obs, done,
+                  discount=None,
                   priorities: Optional[torch.Tensor] = None) -> torch.Tensor:
-        idx = super().add_batch(obs, action, reward, next_obs, done)
+        idx = super().add_batch(obs, action, reward, next_obs, done,
+                                discount=discount)
         if priorities is None:
             p = torch.full((idx.numel(),), self.max_priority ** self.alpha,
                            device=self.device)

@@ -18,21 +18,30 @@ def _init_linear(m: nn.Module) -> None:
 
 
 class QNet(nn.Module):
-    """obs → Q-values; optional dueling decomposition."""
+    """obs → Q-values; optional dueling decomposition and NoisyNet heads
+    (noisy layers per Fortunato et al. 2018 — declared but unimplemented in
+    the reference, rl_args.py:163-315)."""
 
     def __init__(self, obs_dim: int, action_dim: int, hidden_dim: int = 128,
-                 dueling: bool = False):
+                 dueling: bool = False, noisy: bool = False):
         super().__init__()
+        from .noisy import NoisyLinear
         self.dueling = dueling
+        self.noisy = noisy
+        lin = NoisyLinear if noisy else nn.Linear
         self.body = nn.Sequential(
             nn.Linear(obs_dim, hidden_dim), nn.ReLU(),
-            nn.Linear(hidden_dim, hidden_dim), nn.ReLU())
+            lin(hidden_dim, hidden_dim), nn.ReLU())
         if dueling:
-            self.value_head = nn.Linear(hidden_dim, 1)
-            self.adv_head = nn.Linear(hidden_dim, action_dim)
+            self.value_head = lin(hidden_dim, 1)
+            self.adv_head = lin(hidden_dim, action_dim)
         else:
-            self.head = nn.Linear(hidden_dim, action_dim)
+            self.head = lin(hidden_dim, action_dim)
         self.apply(_init_linear)
+
+    def reset_noise(self):
+        from .noisy import reset_noise
+        reset_noise(self)
 
     def forward(self, obs: torch.Tensor) -> torch.Tensor:
         h = self.body(obs)

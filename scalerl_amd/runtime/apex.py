@@ -161,10 +161,16 @@ class ApexTrainer:
         self.flat = FlatParams(self.model, device=dev)
         self.target_flat = FlatParams(self.target_model, device=dev)
         self.optimizer = FusedAdam(self.flat.flat, lr=args.learning_rate)
-        self.buffer = PrioritizedReplayBuffer(
-            args.buffer_size, self.obs_shape, alpha=args.per_alpha,
-            obs_dtype=torch.uint8, device=dev, gamma=args.gamma,
-            seed=args.seed)
+        if args.use_per:
+            self.buffer = PrioritizedReplayBuffer(
+                args.buffer_size, self.obs_shape, alpha=args.per_alpha,
+                obs_dtype=torch.uint8, device=dev, gamma=args.gamma,
+                seed=args.seed)
+        else:  # ParallelDQN mode: same topology, uniform replay
+            from ..data import ReplayBuffer
+            self.buffer = ReplayBuffer(
+                args.buffer_size, self.obs_shape, obs_dtype=torch.uint8,
+                device=dev, gamma=args.gamma, seed=args.seed)
         self.beta_per = args.per_beta
         if dev.type == "cuda":
             from ..parallel.rollout import PinRegistry
@@ -202,21 +208,25 @@ class ApexTrainer:
         flat_d = fd.reshape(-1)
         flat_disc = disc.reshape(-1)
 
-        # initial priorities: |TD| under the current nets (batched)
-        q = self.model(flat_obs).gather(1, flat_act.unsqueeze(1)).squeeze(1)
-        qn_t = self.target_model(next_obs)
-        if args.double_dqn:
-            astar = self.model(next_obs).argmax(dim=1, keepdim=True)
+        if args.use_per:
+            # initial priorities: |TD| under the current nets (batched)
+            q = self.model(flat_obs).gather(1, flat_act.unsqueeze(1)).squeeze(1)
+            qn_t = self.target_model(next_obs)
+            if args.double_dqn:
+                astar = self.model(next_obs).argmax(dim=1, keepdim=True)
+            else:
+                astar = qn_t.argmax(dim=1, keepdim=True)
+            target = flat_r + flat_disc * qn_t.gather(1, astar).squeeze(1)
+            prio = (q - target).abs() + 1e-6
+            self.buffer.add_batch(flat_obs, flat_act, flat_r, next_obs,
+                                  flat_d, discount=flat_disc,
+                                  priorities=prio)
+            # keep max_priority tracking without a sync storm
+            self.buffer.max_priority = max(self.buffer.max_priority,
+                                           float(prio.max()))
         else:
-            astar = qn_t.argmax(dim=1, keepdim=True)
-        target = flat_r + flat_disc * qn_t.gather(1, astar).squeeze(1)
-        prio = (q - target).abs() + 1e-6
-
-        self.buffer.add_batch(flat_obs, flat_act, flat_r, next_obs, flat_d,
-                              priorities=prio)
-        # keep max_priority tracking without a sync storm
-        self.buffer.max_priority = max(self.buffer.max_priority,
-                                       float(prio.max()))
+            self.buffer.add_batch(flat_obs, flat_act, flat_r, next_obs,
+                                  flat_d, discount=flat_disc)
         return flat_obs.shape[0]
 
     def train_iteration(self) -> Dict[str, float]:
@@ -238,8 +248,17 @@ class ApexTrainer:
             return stats
         # 2) SGD steps
         for _ in range(args.learner_update_times):
-            batch, idx, prio, p_total, p_min = \
-                self.buffer.sample_with_priorities(args.batch_size)
+            if args.use_per:
+                batch, idx, prio, p_total, p_min = \
+                    self.buffer.sample_with_priorities(args.batch_size)
+                per_kw = dict(
+                    prios=prio,
+                    p_total=p_total.reshape(1) if p_total.dim() == 0 else p_total,
+                    p_min=p_min.reshape(1) if p_min.dim() == 0 else p_min,
+                    beta=self.beta_per, replay_size=len(self.buffer))
+            else:
+                batch = self.buffer.sample(args.batch_size)
+                per_kw = {}
             self.flat.flat_grad.zero_()
             q = self.model(batch["obs"])
             with torch.no_grad():
@@ -247,16 +266,14 @@ class ApexTrainer:
                 qn_o = self.model(batch["next_obs"]) if args.double_dqn else None
             loss, td_abs = fused_td_loss(
                 q, qn_o, qn_t, batch["action"], batch["reward"],
-                batch["discount"], prios=prio,
-                p_total=p_total.reshape(1) if p_total.dim() == 0 else p_total,
-                p_min=p_min.reshape(1) if p_min.dim() == 0 else p_min,
-                beta=self.beta_per, replay_size=len(self.buffer))
+                batch["discount"], **per_kw)
             loss.backward()
             all_reduce_flat(self.flat.flat_grad, average=True)
             if args.max_grad_norm > 0:
                 clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
             self.optimizer.step(self.flat.flat_grad)
-            self.buffer.update_priorities(idx, td_abs)
+            if args.use_per:
+                self.buffer.update_priorities(idx, td_abs)
             self.learn_iters += 1
             stats["loss"] = loss
             # β anneal → 1
@@ -292,3 +309,21 @@ class ApexTrainer:
                 torch.cuda.synchronize()
             self._pins.unpin_all()
             self._pins = None
+
+
+class ParallelDQNTrainer(ApexTrainer):
+    """Self-contained actor-learner DQN (reference parallel_dqn.py:106-443
+    semantics on the shared runtime): N actor processes with a common
+    ε-greedy schedule feed transition chunks to a central learner with a
+    UNIFORM replay buffer; periodic hard target sync and weight
+    publication.  Identical topology to Ape-X minus prioritization — pass
+    an ApexArguments with ``use_per=False`` (the constructor enforces it)."""
+
+    def __init__(self, args: ApexArguments, device: Optional[str] = None):
+        args.use_per = False
+        super().__init__(args, device=device)
+
+    def actor_eps(self, i: int) -> float:
+        # single shared schedule endpoint (parallel_dqn.py eps handling),
+        # not Ape-X's per-actor spread
+        return self.args.eps_greedy_end

@@ -36,10 +36,21 @@ class DQNAgent(BaseAgent):
         super().__init__(args)
         self.device = torch.device(device)
         self.action_dim = action_dim
-        self.model = QNet(obs_dim, action_dim, args.hidden_dim,
-                          dueling=args.dueling_dqn).to(self.device)
-        self.target_model = QNet(obs_dim, action_dim, args.hidden_dim,
-                                 dueling=args.dueling_dqn).to(self.device)
+        self.categorical = args.categorical_dqn
+        self.noisy = args.noisy_dqn
+
+        def build():
+            if self.categorical:
+                from ..models.noisy import CategoricalQNet
+                return CategoricalQNet(obs_dim, action_dim, args.hidden_dim,
+                                       num_atoms=args.num_atoms,
+                                       v_min=args.v_min, v_max=args.v_max,
+                                       noisy=self.noisy)
+            return QNet(obs_dim, action_dim, args.hidden_dim,
+                        dueling=args.dueling_dqn, noisy=self.noisy)
+
+        self.model = build().to(self.device)
+        self.target_model = build().to(self.device)
         self.target_model.load_state_dict(self.model.state_dict())
         self.flat = FlatParams(self.model, device=self.device)
         self.target_flat = FlatParams(self.target_model, device=self.device)
@@ -52,9 +63,16 @@ class DQNAgent(BaseAgent):
 
     @torch.no_grad()
     def get_action(self, obs: np.ndarray) -> np.ndarray:
-        """ε-greedy batch action (dqn_agent.py:90-112)."""
+        """ε-greedy batch action (dqn_agent.py:90-112); with NoisyNet the
+        exploration comes from resampled parameter noise instead."""
         obs = np.atleast_2d(obs)
         n = obs.shape[0]
+        if self.noisy:
+            self.model.train()
+            from ..models.noisy import reset_noise
+            reset_noise(self.model)
+            t = torch.as_tensor(obs, dtype=torch.float32, device=self.device)
+            return self.model(t).argmax(dim=-1).cpu().numpy()
         self.eps = self.eps_sched.step(n)
         greedy = self.predict(obs)
         explore = self.rng.random(n) < self.eps
@@ -63,6 +81,7 @@ class DQNAgent(BaseAgent):
 
     @torch.no_grad()
     def predict(self, obs: np.ndarray) -> np.ndarray:
+        self.model.eval()
         t = torch.as_tensor(np.atleast_2d(obs), dtype=torch.float32,
                             device=self.device)
         return self.model(t).argmax(dim=-1).cpu().numpy()
@@ -77,19 +96,33 @@ class DQNAgent(BaseAgent):
         discounts = batch["discount"].to(self.device)
 
         self.flat.flat_grad.zero_()
-        q = self.model(obs)
-        with torch.no_grad():
-            q_next_target = self.target_model(next_obs)
-            q_next_online = self.model(next_obs) if args.double_dqn else None
+        if not self.categorical:
+            self.model.train()
+            q = self.model(obs)
+            with torch.no_grad():
+                q_next_target = self.target_model(next_obs)
+                q_next_online = self.model(next_obs) if args.double_dqn else None
 
         prios = batch.get("priorities")
         p_total = p_min = None
         if prios is not None and hasattr(self, "_per_stats"):
             p_total, p_min = self._per_stats
-        loss, td_abs = fused_td_loss(
-            q, q_next_online, q_next_target, actions, rewards, discounts,
-            prios=prios, p_total=p_total, p_min=p_min, beta=self.per_beta,
-            replay_size=replay_size, huber=False)
+        if self.categorical:
+            from ..models.noisy import c51_loss
+            from ..ops import per_is_weights
+            weights = None
+            if prios is not None:
+                weights = per_is_weights(prios, p_total, p_min, replay_size,
+                                         self.per_beta)
+            self.model.train()
+            loss, td_abs = c51_loss(self.model, self.target_model, obs,
+                                    actions, rewards, discounts, next_obs,
+                                    double=args.double_dqn, weights=weights)
+        else:
+            loss, td_abs = fused_td_loss(
+                q, q_next_online, q_next_target, actions, rewards, discounts,
+                prios=prios, p_total=p_total, p_min=p_min, beta=self.per_beta,
+                replay_size=replay_size, huber=False)
         loss.backward()
         all_reduce_flat(self.flat.flat_grad, average=True)
         if args.max_grad_norm > 0:
