@@ -137,7 +137,7 @@ class DQNAgent(BaseAgent):
         elif self.global_update_step % args.target_update_frequency == 0:
             self.target_flat.flat.copy_(self.flat.flat)
 
-        return {"loss": loss, "td_abs": td_abs, "eps": self.eps}
+        return {"loss": loss.detach(), "td_abs": td_abs, "eps": self.eps}
 
     def set_per_stats(self, p_total, p_min) -> None:
         self._per_stats = (p_total, p_min)
