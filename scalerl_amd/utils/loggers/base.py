@@ -21,14 +21,15 @@ class BaseLogger(abc.ABC):
         self.train_interval = train_interval
         self.test_interval = test_interval
         self.update_interval = update_interval
-        self._last = {"train": -1, "test": -1, "update": -1}
+        self._last = {"train": None, "test": None, "update": None}
 
     @abc.abstractmethod
     def write(self, namespace: str, step: int, data: Dict[str, Any]) -> None:
         ...
 
     def _gated(self, ns: str, step: int, data: Dict[str, Any], interval: int):
-        if step - self._last[ns] >= interval:
+        last = self._last[ns]
+        if last is None or step - last >= interval:
             self.write(ns, step, data)
             self._last[ns] = step
 
