@@ -1,0 +1,84 @@
+"""EXPERIMENTAL: hand-written MFMA Atari conv encoder (forward).
+
+Not yet wired into the default model paths (MIOpen via torch remains the
+production conv until these are validated on hardware — the gpu tests for
+this module are gated behind SCALERL_EXPERIMENTAL=1 so the round-end suite
+stays green).  `mfma_selftest` validates the fragment-layout constants
+first; the convs share them.
+"""
+
+from __future__ import annotations
+
+import ctypes
+
+import torch
+
+from . import _backend
+
+_c = ctypes.c_void_p
+
+
+def _declare_conv(lib):
+    if getattr(lib, "_conv_declared", False):
+        return lib
+    c = ctypes
+    for name in ("atari_conv1_fwd_u8", "atari_conv1_fwd_bf16",
+                 "atari_conv2_fwd", "atari_conv3_fwd"):
+        fn = getattr(lib, name)
+        fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p,
+                       c.c_long, c.c_int, c.c_void_p]
+        fn.restype = c.c_int
+    lib.mfma_selftest.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p,
+                                  c.c_void_p]
+    lib.mfma_selftest.restype = c.c_int
+    lib._conv_declared = True
+    return lib
+
+
+def mfma_selftest() -> bool:
+    """Validate the 16x16x32 bf16 fragment maps on-device."""
+    lib = _declare_conv(_backend.lib())
+    g = torch.Generator().manual_seed(0)
+    A = torch.randn(16, 32, generator=g).cuda()
+    B = torch.randn(32, 16, generator=g).cuda()  # asymmetric (guide G9)
+    D = torch.empty(16, 16, device="cuda")
+    ret = lib.mfma_selftest(_c(A.data_ptr()), _c(B.data_ptr()),
+                            _c(D.data_ptr()), _backend.current_stream())
+    _backend.check(ret, "mfma_selftest")
+    want = (A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float())
+    return torch.allclose(D.cpu(), want.cpu(), rtol=1e-2, atol=1e-2)
+
+
+_SHAPES = {
+    1: ("atari_conv1_fwd", (4, 84, 84), (32, 20, 20)),
+    2: ("atari_conv2_fwd", (32, 20, 20), (64, 9, 9)),
+    3: ("atari_conv3_fwd", (64, 9, 9), (64, 7, 7)),
+}
+
+
+@torch.no_grad()
+def atari_conv_fwd(layer: int, x: torch.Tensor, weight: torch.Tensor,
+                   bias: torch.Tensor = None, relu: bool = True) -> torch.Tensor:
+    """Forward one encoder conv.  layer 1 accepts uint8 (fused /255) or
+    bf16; layers 2-3 take bf16.  Returns bf16 [N, K, OH, OW]."""
+    name, in_shape, out_shape = _SHAPES[layer]
+    assert tuple(x.shape[1:]) == in_shape, (x.shape, in_shape)
+    lib = _declare_conv(_backend.lib())
+    if layer == 1:
+        fn = (lib.atari_conv1_fwd_u8 if x.dtype == torch.uint8
+              else lib.atari_conv1_fwd_bf16)
+        if x.dtype != torch.uint8:
+            x = x.to(torch.bfloat16)
+    else:
+        fn = getattr(lib, name)
+        x = x.to(torch.bfloat16)
+    w = weight.to(torch.bfloat16).contiguous()
+    b = bias.float().contiguous() if bias is not None else None
+    out = torch.empty((x.shape[0], *out_shape), dtype=torch.bfloat16,
+                      device=x.device)
+    ret = fn(_c(x.contiguous().data_ptr()), _c(w.data_ptr()),
+             _c(b.data_ptr()) if b is not None else None,
+             _c(out.data_ptr()), x.shape[0], int(relu),
+             _backend.current_stream())
+    _backend.check(ret, name)
+    return out
