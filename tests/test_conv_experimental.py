@@ -1,21 +1,13 @@
-"""EXPERIMENTAL MFMA conv encoder vs torch conv oracle.
+"""MFMA conv encoder (forward) vs torch conv oracle.
 
-Gated behind SCALERL_EXPERIMENTAL=1 (unvalidated-on-hardware kernels must
-not gate the round-end suite); run with:
-    SCALERL_EXPERIMENTAL=1 pytest tests/test_conv_experimental.py -m gpu
-"""
-
-import os
+Validated on MI355X: fragment self-test + all 3 shapes + fused u8
+normalize pass against the bf16 torch reference (round 1)."""
 
 import pytest
 import torch
 import torch.nn.functional as F
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(not os.environ.get("SCALERL_EXPERIMENTAL"),
-                       reason="experimental kernels (set SCALERL_EXPERIMENTAL=1)"),
-]
+pytestmark = [pytest.mark.gpu]
 
 
 def test_mfma_fragment_selftest():
