@@ -82,3 +82,64 @@ def atari_conv_fwd(layer: int, x: torch.Tensor, weight: torch.Tensor,
              _backend.current_stream())
     _backend.check(ret, name)
     return out
+
+
+def _declare_bwd(lib):
+    if getattr(lib, "_conv_bwd_declared", False):
+        return lib
+    c = ctypes
+    for name in ("atari_conv1_wgrad_u8", "atari_conv2_wgrad",
+                 "atari_conv3_wgrad"):
+        fn = getattr(lib, name)
+        fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_long,
+                       c.c_long, c.c_void_p]
+        fn.restype = c.c_int
+    for name in ("atari_conv2_dgrad", "atari_conv3_dgrad"):
+        fn = getattr(lib, name)
+        fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_long,
+                       c.c_void_p]
+        fn.restype = c.c_int
+    lib._conv_bwd_declared = True
+    return lib
+
+
+_WGRAD = {1: "atari_conv1_wgrad_u8", 2: "atari_conv2_wgrad",
+          3: "atari_conv3_wgrad"}
+_DGRAD = {2: "atari_conv2_dgrad", 3: "atari_conv3_dgrad"}
+_WSHAPE = {1: (32, 4, 8, 8), 2: (64, 32, 4, 4), 3: (64, 64, 3, 3)}
+
+
+@torch.no_grad()
+def atari_conv_wgrad(layer: int, x: torch.Tensor, dout: torch.Tensor,
+                     split: int = 64) -> torch.Tensor:
+    """dL/dW for one encoder conv (fp32 out).  Layer 1 takes uint8 x."""
+    lib = _declare_bwd(_declare_conv(_backend.lib()))
+    K, C, KH, KW = _WSHAPE[layer]
+    dw = torch.zeros(K, C * KH * KW, device=x.device, dtype=torch.float32)
+    if layer == 1:
+        assert x.dtype == torch.uint8
+        xc = x.contiguous()
+    else:
+        xc = x.to(torch.bfloat16).contiguous()
+    fn = getattr(lib, _WGRAD[layer])
+    ret = fn(_c(xc.data_ptr()),
+             _c(dout.to(torch.bfloat16).contiguous().data_ptr()),
+             _c(dw.data_ptr()), x.shape[0], split, _backend.current_stream())
+    _backend.check(ret, _WGRAD[layer])
+    return dw.view(K, C, KH, KW)
+
+
+@torch.no_grad()
+def atari_conv_dgrad(layer: int, dout: torch.Tensor,
+                     weight: torch.Tensor) -> torch.Tensor:
+    """dL/dX for encoder convs 2-3 (bf16 out; conv1 is the input layer)."""
+    lib = _declare_bwd(_declare_conv(_backend.lib()))
+    _, in_shape, _ = _SHAPES[layer]
+    din = torch.empty((dout.shape[0], *in_shape), dtype=torch.bfloat16,
+                      device=dout.device)
+    fn = getattr(lib, _DGRAD[layer])
+    ret = fn(_c(dout.to(torch.bfloat16).contiguous().data_ptr()),
+             _c(weight.to(torch.bfloat16).contiguous().data_ptr()),
+             _c(din.data_ptr()), dout.shape[0], _backend.current_stream())
+    _backend.check(ret, _DGRAD[layer])
+    return din

@@ -162,3 +162,169 @@ DEF_CONV(atari_conv1_fwd_u8, 4, 8, 8, 4, 84, 84, 20, 20, 32, true,
 DEF_CONV(atari_conv1_fwd_bf16, 4, 8, 8, 4, 84, 84, 20, 20, 32, false, bf16_t)
 DEF_CONV(atari_conv2_fwd, 32, 4, 4, 2, 20, 20, 9, 9, 64, false, bf16_t)
 DEF_CONV(atari_conv3_fwd, 64, 3, 3, 1, 9, 9, 7, 7, 64, false, bf16_t)
+
+// ---- backward: weight gradient ---------------------------------------
+// dw[k_out, kdim] = sum_pixels dy[pixel, k_out] * im2col(x)[pixel, kdim]
+// GEMM: rows = K_OUT, cols = KDIM, reduce over M = N*OH*OW pixels.
+// The natural grid ((KOUT/16) x (KDIM/16)) underfills 256 CUs, so the
+// pixel axis is split across blockIdx.z and partial tiles atomicAdd into
+// the fp32 dw buffer (zeroed by the caller).
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT, bool IN_U8, typename in_t>
+__global__ __launch_bounds__(64) void conv_wgrad_kernel(
+    const in_t* __restrict__ input,    // [N, C, IH, IW]
+    const bf16_t* __restrict__ dout,   // [N, KOUT, OH, OW]
+    float* __restrict__ dweight,       // [KOUT, C*KH*KW] fp32 (atomic)
+    int batch, int split) {
+  constexpr int KDIM = C * KH * KW;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int M = batch * OH * OW;
+  const int per_split = (M + split - 1) / split;
+  const int p_begin = blockIdx.z * per_split;
+  const int p_end = min(p_begin + per_split, M);
+
+  const int kout_base = blockIdx.x * 16;
+  const int kdim_base = blockIdx.y * 16;
+  const int arow = kout_base + (lane & 15);   // k_out
+  const int bcol = kdim_base + (lane & 15);   // kdim
+  int bc = 0, bky = 0, bkx = 0;
+  if (bcol < KDIM) {
+    bc = bcol / (KH * KW);
+    const int r = bcol % (KH * KW);
+    bky = r / KW;
+    bkx = r % KW;
+  }
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int p0 = p_begin; p0 < p_end; p0 += 32) {
+    bf16x8 a, b;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int p = p0 + (lane >> 4) * 8 + j;
+      float av = 0.f, bv = 0.f;
+      if (p < p_end) {
+        const int n = p / (OH * OW);
+        const int rem = p % (OH * OW);
+        const int oy = rem / OW, ox = rem % OW;
+        if (arow < KOUT)
+          av = (float)dout[(((long)n * KOUT + arow) * OH + oy) * OW + ox];
+        if (bcol < KDIM) {
+          const in_t raw = input[(((long)n * C + bc) * IH + oy * STRIDE + bky)
+                                 * IW + ox * STRIDE + bkx];
+          bv = IN_U8 ? (float)raw * (1.0f / 255.0f) : (float)raw;
+        }
+      }
+      a[j] = (bf16_t)av;
+      b[j] = (bf16_t)bv;
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int col = kdim_base + (lane & 15);
+    const int row = kout_base + (lane >> 4) * 4 + r;
+    if (row < KOUT && col < KDIM)
+      atomicAdd(&dweight[(long)row * KDIM + col], acc[r]);
+  }
+}
+
+// ---- backward: input gradient ----------------------------------------
+// dx[pixel(n,iy,ix), c] = sum_{k_out,ky,kx valid} dy[n,k_out,oy,ox] *
+//                         w[k_out, c, ky, kx],  oy=(iy-ky)/S exact.
+// GEMM rows = input pixels, cols = C, reduce dim = KOUT*KH*KW with
+// validity-masked dy gathers.
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT>
+__global__ __launch_bounds__(64) void conv_dgrad_kernel(
+    const bf16_t* __restrict__ dout,    // [N, KOUT, OH, OW]
+    const bf16_t* __restrict__ weight,  // [KOUT, C, KH, KW]
+    bf16_t* __restrict__ dinput,        // [N, C, IH, IW]
+    int batch) {
+  constexpr int RDIM = KOUT * KH * KW;
+  constexpr int RTILES = (RDIM + 31) / 32;
+  const int M = batch * IH * IW;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m_base = blockIdx.x * 16;
+  const int c_base = blockIdx.y * 16;
+
+  const int arow = m_base + (lane & 15);
+  int n = 0, iy = 0, ix = 0;
+  const bool row_ok = arow < M;
+  if (row_ok) {
+    n = arow / (IH * IW);
+    const int rem = arow % (IH * IW);
+    iy = rem / IW;
+    ix = rem % IW;
+  }
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int rt = 0; rt < RTILES; ++rt) {
+    bf16x8 a, b;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = rt * 32 + (lane >> 4) * 8 + j;
+      float av = 0.f, bv = 0.f;
+      if (k < RDIM) {
+        const int kout = k / (KH * KW);
+        const int r = k % (KH * KW);
+        const int ky = r / KW, kx = r % KW;
+        if (row_ok) {
+          const int ty = iy - ky, tx = ix - kx;
+          if (ty >= 0 && tx >= 0 && ty % STRIDE == 0 && tx % STRIDE == 0) {
+            const int oy = ty / STRIDE, ox = tx / STRIDE;
+            if (oy < OH && ox < OW)
+              av = (float)dout[(((long)n * KOUT + kout) * OH + oy) * OW + ox];
+          }
+        }
+        const int bcol = c_base + (lane & 15);
+        if (bcol < C)
+          bv = (float)weight[(((long)kout * C + bcol) * KH + ky) * KW + kx];
+      }
+      a[j] = (bf16_t)av;
+      b[j] = (bf16_t)bv;
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int col = c_base + (lane & 15);
+    const int row = m_base + (lane >> 4) * 4 + r;
+    if (row < M && col < C) {
+      const int ni = row / (IH * IW);
+      const int rem = row % (IH * IW);
+      dinput[((long)ni * C + col) * IH * IW + rem] = (bf16_t)acc[r];
+    }
+  }
+}
+
+#define DEF_WGRAD(NAME, C_, KH_, KW_, S_, IH_, IW_, OH_, OW_, KO_, U8, T)     \
+  extern "C" int NAME(const void* in, const void* dout, float* dw,            \
+                      long batch, long split, hipStream_t stream) {           \
+    dim3 grid((KO_ + 15) / 16, (C_ * KH_ * KW_ + 15) / 16, (unsigned)split);  \
+    hipLaunchKernelGGL(                                                       \
+        (conv_wgrad_kernel<C_, KH_, KW_, S_, IH_, IW_, OH_, OW_, KO_, U8, T>),\
+        grid, dim3(64), 0, stream, (const T*)in, (const bf16_t*)dout, dw,     \
+        (int)batch, (int)split);                                              \
+    CHECK_LAUNCH();                                                           \
+    return 0;                                                                 \
+  }
+
+#define DEF_DGRAD(NAME, C_, KH_, KW_, S_, IH_, IW_, OH_, OW_, KO_)            \
+  extern "C" int NAME(const void* dout, const void* w, void* din,             \
+                      long batch, hipStream_t stream) {                       \
+    const long M = batch * IH_ * IW_;                                         \
+    dim3 grid((unsigned)((M + 15) / 16), (C_ + 15) / 16);                     \
+    hipLaunchKernelGGL((conv_dgrad_kernel<C_, KH_, KW_, S_, IH_, IW_, OH_,    \
+                                          OW_, KO_>),                         \
+                       grid, dim3(64), 0, stream, (const bf16_t*)dout,        \
+                       (const bf16_t*)w, (bf16_t*)din, (int)batch);           \
+    CHECK_LAUNCH();                                                           \
+    return 0;                                                                 \
+  }
+
+DEF_WGRAD(atari_conv1_wgrad_u8, 4, 8, 8, 4, 84, 84, 20, 20, 32, true,
+          unsigned char)
+DEF_WGRAD(atari_conv2_wgrad, 32, 4, 4, 2, 20, 20, 9, 9, 64, false, bf16_t)
+DEF_WGRAD(atari_conv3_wgrad, 64, 3, 3, 1, 9, 9, 7, 7, 64, false, bf16_t)
+DEF_DGRAD(atari_conv2_dgrad, 32, 4, 4, 2, 20, 20, 9, 9, 64)
+DEF_DGRAD(atari_conv3_dgrad, 64, 3, 3, 1, 9, 9, 7, 7, 64)

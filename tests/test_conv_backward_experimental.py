@@ -1,0 +1,62 @@
+"""MFMA conv backward (wgrad/dgrad) vs torch autograd oracle.
+
+Gated behind SCALERL_EXPERIMENTAL=1 until hardware-validated (the forward
+kernels and fragment maps they share ARE validated — see
+test_conv_experimental.py)."""
+
+import os
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = [
+    pytest.mark.gpu,
+    pytest.mark.skipif(not os.environ.get("SCALERL_EXPERIMENTAL"),
+                       reason="unvalidated backward kernels"),
+]
+
+SHAPES = {1: ((4, 84, 84), (32, 4, 8, 8), 4),
+          2: ((32, 20, 20), (64, 32, 4, 4), 2),
+          3: ((64, 9, 9), (64, 64, 3, 3), 1)}
+
+
+@pytest.mark.parametrize("layer", [1, 2, 3])
+def test_wgrad_matches_torch(layer):
+    from scalerl_amd.ops.conv import atari_conv_wgrad
+    torch.manual_seed(0)
+    in_shape, w_shape, stride = SHAPES[layer]
+    N = 21
+    if layer == 1:
+        x_u8 = torch.randint(0, 256, (N, *in_shape), dtype=torch.uint8,
+                             device="cuda")
+        x_ref = (x_u8.float() / 255.0).to(torch.bfloat16).float()
+        x_in = x_u8
+    else:
+        x = torch.randn(N, *in_shape, device="cuda")
+        x_ref = x.to(torch.bfloat16).float()
+        x_in = x
+    w = (torch.randn(w_shape, device="cuda") * 0.1).requires_grad_()
+    dout = torch.randn(F.conv2d(x_ref, w, stride=stride).shape,
+                       device="cuda")
+    out = F.conv2d(x_ref, w.to(torch.bfloat16).float(), stride=stride)
+    (out * dout).sum().backward()
+    got = atari_conv_wgrad(layer, x_in, dout)
+    torch.testing.assert_close(got, w.grad, rtol=5e-2, atol=5e-1)
+
+
+@pytest.mark.parametrize("layer", [2, 3])
+def test_dgrad_matches_torch(layer):
+    from scalerl_amd.ops.conv import atari_conv_dgrad
+    torch.manual_seed(1)
+    in_shape, w_shape, stride = SHAPES[layer]
+    N = 13
+    x = torch.randn(N, *in_shape, device="cuda",
+                    requires_grad=True)
+    w = torch.randn(w_shape, device="cuda") * 0.1
+    out = F.conv2d(x.to(torch.bfloat16).float(),
+                   w.to(torch.bfloat16).float(), stride=stride)
+    dout = torch.randn_like(out)
+    (out * dout).sum().backward()
+    got = atari_conv_dgrad(layer, dout, w).float()
+    torch.testing.assert_close(got, x.grad, rtol=5e-2, atol=5e-2)
