@@ -28,11 +28,18 @@ from ..ops import MaskedLSTM
 
 class AtariNet(nn.Module):
     def __init__(self, observation_shape=(4, 84, 84), num_actions: int = 6,
-                 use_lstm: bool = True):
+                 use_lstm: bool = True, native_conv: bool = None):
         super().__init__()
         self.observation_shape = tuple(observation_shape)
         self.num_actions = num_actions
         self.use_lstm = use_lstm
+        # EXPERIMENTAL: hand-written MFMA encoder convs (ops/conv.py);
+        # requires the standard (4,84,84) shape; default off until the
+        # backward kernels are hardware-validated.
+        if native_conv is None:
+            import os
+            native_conv = bool(os.environ.get("SCALERL_NATIVE_CONV"))
+        self.native_conv = native_conv and tuple(observation_shape) == (4, 84, 84)
 
         c = observation_shape[0]
         self.conv1 = nn.Conv2d(c, 32, kernel_size=8, stride=4)
@@ -55,6 +62,12 @@ class AtariNet(nn.Module):
 
     def encode(self, x: torch.Tensor) -> torch.Tensor:
         """uint8 [N,C,H,W] → fp feature [N,512]."""
+        if self.native_conv and x.is_cuda:
+            from ..ops.conv import native_conv
+            h = native_conv(1, x, self.conv1.weight, self.conv1.bias)
+            h = native_conv(2, h, self.conv2.weight, self.conv2.bias)
+            h = native_conv(3, h, self.conv3.weight, self.conv3.bias)
+            return F.relu(self.fc(torch.flatten(h, 1)))
         x = x.float() / 255.0
         x = F.relu(self.conv1(x))
         x = F.relu(self.conv2(x))

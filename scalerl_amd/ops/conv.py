@@ -143,3 +143,36 @@ def atari_conv_dgrad(layer: int, dout: torch.Tensor,
              _c(din.data_ptr()), dout.shape[0], _backend.current_stream())
     _backend.check(ret, _DGRAD[layer])
     return din
+
+
+class _NativeConvFn(torch.autograd.Function):
+    """Autograd over the native encoder convs (layer 2/3: full backward;
+    layer 1: wgrad only — it is the input layer)."""
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda")
+    def forward(ctx, x, weight, bias, layer, relu):
+        out = atari_conv_fwd(layer, x, weight, bias, relu=relu)
+        ctx.save_for_backward(x, weight, out)
+        ctx.layer, ctx.relu = layer, relu
+        return out
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, dout):
+        x, weight, out = ctx.saved_tensors
+        layer = ctx.layer
+        if ctx.relu:
+            dout = dout * (out > 0)
+        dw = atari_conv_wgrad(layer, x, dout)
+        db = dout.float().sum(dim=(0, 2, 3))
+        dx = None
+        if layer in _DGRAD and ctx.needs_input_grad[0]:
+            dx = atari_conv_dgrad(layer, dout, weight)
+        return dx, dw, db, None, None
+
+
+def native_conv(layer: int, x, weight, bias=None, relu: bool = True):
+    """Differentiable native encoder conv (EXPERIMENTAL, opt-in via
+    AtariNet(native_conv=True) / SCALERL_NATIVE_CONV=1)."""
+    return _NativeConvFn.apply(x, weight, bias, layer, relu)

@@ -71,8 +71,11 @@ class ImpalaTrainer:
             f"batch_size ({args.batch_size}) must be a multiple of " \
             f"envs_per_actor ({E})"
         self.slots_per_batch = args.batch_size // E
+        # enough slots that every actor can have one in flight plus two
+        # batches queued; 2x actors would double host shm for no throughput
+        # (8 ranks x 24 actors x 290 MB slots must fit /dev/shm)
         num_buffers = args.num_buffers or (
-            2 * args.num_actors + 2 * self.slots_per_batch)
+            args.num_actors + 2 * self.slots_per_batch + 2)
 
         # ---- CPU phase: shared actor model + store + actor processes ----
         # (everything here must precede any CUDA/HIP initialization)
