@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import enum
 import multiprocessing as mp
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, Optional, Sequence
 
 import numpy as np
 import torch

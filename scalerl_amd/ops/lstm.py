@@ -17,7 +17,7 @@ checkpoints interop with torch LSTMs.
 from __future__ import annotations
 
 import ctypes
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 import torch.nn as nn

@@ -8,7 +8,7 @@ apex/worker.py:134-161 (IS weights, priority update).
 from __future__ import annotations
 
 import ctypes
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

@@ -11,8 +11,6 @@ from __future__ import annotations
 
 import collections
 import ctypes
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 

@@ -22,12 +22,10 @@ actors→node, both as raw tensor frames.
 
 from __future__ import annotations
 
-import io
 import pickle
 import socket
 import struct
 import threading
-import time
 from typing import Any, Callable, Dict, List, Optional, Tuple
 
 import torch

@@ -25,14 +25,11 @@ kernels (separate HIP contexts timeslice the CU array).
 
 from __future__ import annotations
 
-import ctypes
 import queue
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
-import numpy as np
 import torch
-import torch.multiprocessing as mp
 
 
 class InferenceSlots:

@@ -29,13 +29,11 @@ from __future__ import annotations
 
 import ctypes
 import ctypes.util
-import os
 import queue
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import numpy as np
 import torch
-import torch.multiprocessing as mp
 
 
 class RolloutStore:

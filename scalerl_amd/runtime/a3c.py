@@ -21,9 +21,7 @@ kernels the GPU learners use; CPU reference path here).
 
 from __future__ import annotations
 
-import os
 import time
-from typing import Dict, Optional
 
 import numpy as np
 import torch

@@ -22,7 +22,6 @@ iteration consumes rollout_length × batch_size env steps.
 
 from __future__ import annotations
 
-import copy
 import os
 import time
 from typing import Dict, List, Optional
@@ -36,8 +35,7 @@ from ..models.atari import AtariNet
 from ..ops import FusedRMSprop, clip_grad_norm_, impala_loss
 from ..parallel import FlatParams, all_reduce_flat, get_rank, get_world_size
 from ..parallel.inference import InferenceSlots, inference_worker
-from ..parallel.rollout import (BatchGatherer, RolloutStore, actor_loop,
-                                pin_store)
+from ..parallel.rollout import BatchGatherer, RolloutStore, actor_loop
 from ..utils import Timings, get_logger
 from ..utils.checkpoint import load_checkpoint, save_checkpoint
 

@@ -18,10 +18,8 @@ implemented fresh per SURVEY.md §7 step 9, following Wijmans et al. 2020:
 
 from __future__ import annotations
 
-import time
 from typing import Dict, Optional
 
-import numpy as np
 import torch
 import torch.distributed as dist
 import torch.nn.functional as F

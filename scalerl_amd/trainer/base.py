@@ -6,7 +6,7 @@ from __future__ import annotations
 
 import os
 import time
-from typing import Dict, Optional
+from typing import Dict
 
 from ..parallel.dist import get_rank
 from ..utils import get_logger

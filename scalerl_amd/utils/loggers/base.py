@@ -10,7 +10,7 @@ importable, keeping the reference's logger surface.
 from __future__ import annotations
 
 import abc
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 
 class BaseLogger(abc.ABC):

@@ -6,7 +6,7 @@ from __future__ import annotations
 import sys
 import time
 from multiprocessing import Pool
-from typing import Callable, Iterable, List, Optional
+from typing import Callable, Iterable, List
 
 
 class ProgressBar:
