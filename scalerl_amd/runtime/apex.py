@@ -116,7 +116,7 @@ class ApexTrainer:
                         else "spawn")
         ctx = mp.get_context(self._mp_ctx)
         self.free_q = ctx.SimpleQueue()
-        self.full_q = ctx.SimpleQueue()
+        self.full_q = ctx.Queue()  # timeout-capable: the learner watchdogs it
         self.stop_event = ctx.Event()
         self.step_counter = ctx.Value("l", 0)
         self.actors: List[mp.Process] = []
@@ -233,11 +233,18 @@ class ApexTrainer:
         # outproduce the ingest forward would live-lock the learner here
         ingested = 0
         max_slots = 4
+        import queue as _queue
         for _ in range(max_slots):
             if self.full_q.empty():
                 if len(self.buffer) >= args.warmup_learn_steps or ingested:
                     break
-            slot = self.full_q.get()
+            try:
+                slot = self.full_q.get(timeout=60.0)
+            except _queue.Empty:
+                dead = [p.name for p in self.actors if not p.is_alive()]
+                if dead:
+                    raise RuntimeError(f"actor process(es) died: {dead}")
+                continue
             ingested += self.ingest_slot(slot)
             self.free_q.put(slot)
             self.global_step += self.chunk_len * args.envs_per_actor

@@ -97,7 +97,7 @@ class ImpalaTrainer:
                         else "spawn")
         ctx = mp.get_context(self._mp_ctx)
         self.free_q = ctx.SimpleQueue()
-        self.full_q = ctx.SimpleQueue()
+        self.full_q = ctx.Queue()  # timeout-capable: the learner watchdogs it
         self.stop_event = ctx.Event()
         self.step_counter = ctx.Value("l", 0)
         self.episode_q = ctx.Queue(maxsize=256)
@@ -230,8 +230,30 @@ class ImpalaTrainer:
                 self.weights_version.value += 1
 
     # -- core step ---------------------------------------------------------
+    def _get_full_slot(self, timeout_s: float = 30.0, max_wait_s: float = 600.0):
+        """full_q.get with a liveness watchdog: a dead actor or inference
+        worker turns a silent eternal hang into a fast, explicit error."""
+        import queue as _queue
+        waited = 0.0
+        while True:
+            try:
+                return self.full_q.get(timeout=timeout_s)
+            except _queue.Empty:
+                waited += timeout_s
+                dead = [p.name for p in self.actors if not p.is_alive()]
+                if dead:
+                    raise RuntimeError(f"actor process(es) died: {dead}")
+                if (self.inference_proc is not None
+                        and not self.inference_proc.is_alive()):
+                    raise RuntimeError("GPU inference worker died")
+                if waited >= max_wait_s:
+                    raise RuntimeError(
+                        f"no rollout slot arrived in {waited:.0f}s "
+                        f"(actors alive but not producing)")
+
     def _start_prefetch(self):
-        slot_ids = [self.full_q.get() for _ in range(self.slots_per_batch)]
+        slot_ids = [self._get_full_slot()
+                    for _ in range(self.slots_per_batch)]
         self.timings.time("dequeue")
         token = self.gatherer.start(slot_ids)
         self.timings.time("gather_start")

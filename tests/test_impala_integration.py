@@ -112,3 +112,21 @@ def test_impala_gpu_inference_end_to_end(tmp_path):
         assert t.inference == "gpu" and t.inference_proc.is_alive()
     finally:
         t.shutdown()
+
+
+def test_impala_dead_actor_raises(tmp_path):
+    """A dead actor must surface as an explicit error, not an eternal hang."""
+    t = ImpalaTrainer(_args(tmp_path, num_actors=1))
+    try:
+        t.start_actors()
+        t.setup_learner()
+        t.train_iteration()
+        for p in t.actors:
+            p.terminate()
+            p.join()
+        # drain whatever was already produced, then expect the watchdog
+        with pytest.raises(RuntimeError, match="actor process"):
+            for _ in range(t._num_buffers + 2):
+                t._get_full_slot(timeout_s=0.5, max_wait_s=5.0)
+    finally:
+        t.shutdown()
