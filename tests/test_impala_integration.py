@@ -130,3 +130,21 @@ def test_impala_dead_actor_raises(tmp_path):
                 t._get_full_slot(timeout_s=0.5, max_wait_s=5.0)
     finally:
         t.shutdown()
+
+
+def test_impala_single_actor_reproducible(tmp_path):
+    """Same seed + one actor → bit-identical loss sequence (deterministic
+    replay mode — SURVEY.md §5 race-detection recommendation)."""
+    def run(seed):
+        t = ImpalaTrainer(_args(tmp_path, num_actors=1, envs_per_actor=8,
+                                seed=seed))
+        try:
+            t.start_actors()
+            t.setup_learner()
+            return [float(t.train_iteration()["total_loss"])
+                    for _ in range(3)]
+        finally:
+            t.shutdown()
+    a, b, c = run(11), run(11), run(12)
+    assert a == b
+    assert a != c
