@@ -13,6 +13,7 @@ from .optim import FusedAdam, FusedRMSprop, clip_grad_norm_, fused_polyak_
 from .per import SumTree
 from .scans import discounted_returns, gae, nstep_fold
 from .td import fused_td_loss, per_is_weights, td_loss_reference
+from .ppo import ppo_fused_loss, ppo_loss_reference
 from .vtrace import (VTraceReturns, action_log_probs, impala_loss,
                      impala_loss_reference, vtrace_from_log_rhos,
                      vtrace_reference)
@@ -23,5 +24,6 @@ __all__ = [
     "discounted_returns", "gae", "nstep_fold", "fused_td_loss",
     "per_is_weights", "td_loss_reference", "VTraceReturns",
     "action_log_probs", "impala_loss", "impala_loss_reference",
-    "vtrace_from_log_rhos", "vtrace_reference",
+    "vtrace_from_log_rhos", "vtrace_reference", "ppo_fused_loss",
+    "ppo_loss_reference",
 ]

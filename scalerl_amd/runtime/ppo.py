@@ -161,30 +161,28 @@ class DDPPOTrainer:
                 state = (start_state[0][:, cols], start_state[1][:, cols])
                 logits, values, _ = self.model(obs[:, cols],
                                                notdone[:, cols], state)
-                logits = logits.float()
-                values = values.float()
-                logp_all = F.log_softmax(logits, dim=-1)
-                logp = logp_all.gather(
-                    2, actions[:, cols].unsqueeze(-1)).squeeze(-1)
-                ratio = torch.exp(logp - old_logp[:, cols])
-                a = adv[:, cols]
-                surr1 = ratio * a
-                surr2 = torch.clamp(ratio, 1 - args.clip_eps,
-                                    1 + args.clip_eps) * a
-                pg_loss = -torch.min(surr1, surr2).mean()
-                v_loss = F.mse_loss(values, ret[:, cols])
-                entropy = -(logp_all.exp() * logp_all).sum(-1).mean()
-                loss = (pg_loss + args.value_loss_coef * v_loss
-                        - args.entropy_coef * entropy)
+                # fused HIP clip loss on GPU (one launch incl. analytic
+                # grads), composed reference on CPU
+                from ..ops import ppo_fused_loss
+                k = cols.numel()
+                loss, comps = ppo_fused_loss(
+                    logits.float().reshape(T * k, -1),
+                    values.float().reshape(T * k),
+                    actions[:, cols].reshape(T * k),
+                    old_logp[:, cols].reshape(T * k),
+                    adv[:, cols].reshape(T * k),
+                    ret[:, cols].reshape(T * k),
+                    clip_eps=args.clip_eps, vcoef=args.value_loss_coef,
+                    ecoef=args.entropy_coef)
                 loss.backward()
                 all_reduce_flat(self.flat.flat_grad, average=True)
                 if args.max_grad_norm > 0:
                     clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
                 self.optimizer.step(self.flat.flat_grad)
                 stats = {"loss": float(loss.detach()),
-                         "pg_loss": float(pg_loss.detach()),
-                         "v_loss": float(v_loss.detach()),
-                         "entropy": float(entropy.detach())}
+                         "pg_loss": float(comps[0]),
+                         "v_loss": float(comps[1]),
+                         "entropy": float(comps[2])}
         return stats
 
     def train_iteration(self) -> Dict[str, float]:

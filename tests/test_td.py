@@ -66,3 +66,56 @@ def test_fused_td_gpu_matches_reference(huber, per):
     torch.testing.assert_close(loss.cpu(), loss_ref, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(tda.cpu(), tda_ref, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(qg.grad.cpu(), qc.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_ppo_loss_reference_grads():
+    from scalerl_amd.ops import ppo_fused_loss
+    g = torch.Generator().manual_seed(0)
+    N, A = 64, 5
+    logits = torch.randn(N, A, generator=g, requires_grad=True)
+    values = torch.randn(N, generator=g, requires_grad=True)
+    actions = torch.randint(0, A, (N,), generator=g)
+    old_logp = torch.log_softmax(torch.randn(N, A, generator=g), -1).gather(
+        1, actions.unsqueeze(1)).squeeze(1)
+    adv = torch.randn(N, generator=g)
+    rets = torch.randn(N, generator=g)
+    total, comps = ppo_fused_loss(logits, values, actions, old_logp, adv,
+                                  rets)
+    total.backward()
+    assert logits.grad.abs().sum() > 0 and values.grad.abs().sum() > 0
+
+
+@pytest.mark.gpu
+def test_ppo_fused_loss_matches_reference():
+    from scalerl_amd.ops import ppo_fused_loss
+    from scalerl_amd.ops.ppo import ppo_loss_reference
+    dev = "cuda:0"
+    g = torch.Generator().manual_seed(1)
+    N, A = 512, 4
+    logits = torch.randn(N, A, generator=g)
+    values = torch.randn(N, generator=g)
+    actions = torch.randint(0, A, (N,), generator=g)
+    old_logp = torch.log_softmax(torch.randn(N, A, generator=g), -1).gather(
+        1, actions.unsqueeze(1)).squeeze(1)
+    adv = torch.randn(N, generator=g)
+    rets = torch.randn(N, generator=g)
+
+    lg = logits.to(dev).requires_grad_()
+    vg = values.to(dev).requires_grad_()
+    total, comps = ppo_fused_loss(lg, vg, actions.to(dev), old_logp.to(dev),
+                                  adv.to(dev), rets.to(dev),
+                                  clip_eps=0.2, vcoef=0.5, ecoef=0.01)
+    total.backward()
+
+    lc = logits.clone().requires_grad_()
+    vc = values.clone().requires_grad_()
+    total_ref, pg, v, ent = ppo_loss_reference(lc, actions, old_logp, adv,
+                                               rets, vc, 0.2, 0.5, 0.01)
+    total_ref.backward()
+    torch.testing.assert_close(total.cpu(), total_ref.detach(), rtol=1e-3,
+                               atol=1e-4)
+    torch.testing.assert_close(comps[0].cpu(), pg, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(comps[1].cpu(), v, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(comps[2].cpu(), ent, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(lg.grad.cpu(), lc.grad, rtol=1e-3, atol=1e-5)
+    torch.testing.assert_close(vg.grad.cpu(), vc.grad, rtol=1e-3, atol=1e-5)
