@@ -336,7 +336,11 @@ class ImpalaTrainer:
         if self.device.type == "cuda":
             # don't overwrite weights while the previous publish D2H reads them
             torch.cuda.current_stream().wait_event(self.publish_event)
-        self.optimizer.step(self.flat.flat_grad)
+        # linear lr decay to min_learning_rate over total_steps (the
+        # reference's monobeast-heritage schedule)
+        frac = min(self.global_step / max(args.total_steps, 1), 1.0)
+        lr = max(args.learning_rate * (1.0 - frac), args.min_learning_rate)
+        self.optimizer.step(self.flat.flat_grad, lr=lr)
         self.timings.time("optimize")
         self._publish_weights()
         self.timings.time("publish")
