@@ -71,7 +71,7 @@ def test_impala_learns_synthetic_reward(tmp_path):
     t = ImpalaTrainer(_args(tmp_path, rollout_length=16, batch_size=16,
                             envs_per_actor=8, num_actors=2, use_lstm=False,
                             entropy_cost=0.02, learning_rate=3e-4,
-                            discounting=0.5))
+                            discounting=0.5, total_steps=1 << 40))
     try:
         t.start_actors()
         t.setup_learner()
