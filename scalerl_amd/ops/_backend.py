@@ -42,7 +42,7 @@ def build(verbose: bool = True, arch: str = "gfx950") -> str:
             return _SO_PATH
     hipcc = os.environ.get("HIPCC", "hipcc")
     cmd = [hipcc, f"--offload-arch={arch}", "-O3", "-std=c++17", "-shared",
-           "-fPIC", "-o", _SO_PATH] + srcs
+           "-fPIC", "-lrocblas", "-o", _SO_PATH] + srcs
     if verbose:
         print("[scalerl_amd.ops] building:", " ".join(cmd), file=sys.stderr)
     subprocess.run(cmd, check=True)
@@ -70,6 +70,8 @@ def _declare(lib: ctypes.CDLL) -> None:
         "per_leaf_min": [P, L, L, P, P],
         "lstm_pointwise_fwd": [P, P, P, P, L, L, P],
         "lstm_pointwise_bwd": [P, P, P, P, P, P, P, L, L, P],
+        "masked_lstm_seq_fwd": [P, P, P, P, P, P, P, P, P, P, L, L, L, P],
+        "masked_lstm_seq_bwd": [P, P, P, P, P, P, P, P, P, P, L, L, L, P],
     }
     for name, argtypes in sigs.items():
         fn = getattr(lib, name)

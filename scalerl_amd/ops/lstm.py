@@ -134,6 +134,83 @@ class _MaskedLSTMFn(torch.autograd.Function):
         return dx, None, dh_carry, dc_carry, dw_ih, dw_hh, db, db
 
 
+class _MaskedLSTMSeqFn(torch.autograd.Function):
+    """One LSTM layer unrolled over T with the loop driven from C++
+    (csrc/lstm_seq.hip): per step one rocBLAS SGEMM + one fused kernel, all
+    enqueued natively — removes the ~20 ms/iter of Python launch overhead
+    the per-step path pays (profiles/README.md).  Numerics identical to
+    :class:`_MaskedLSTMFn`.  EXPERIMENTAL: enabled via SCALERL_LSTM_SEQ=1
+    until hardware-validated (its oracle test is test_lstm_seq*)."""
+
+    @staticmethod
+    @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.float32)
+    def forward(ctx, x, notdone, h0, c0, w_ih, w_hh, b_ih, b_hh):
+        lib = _backend.lib()
+        T, B, I = x.shape
+        H = w_hh.shape[1]
+        dev = x.device
+        xg = torch.addmm(b_ih + b_hh, x.reshape(T * B, I),
+                         w_ih.t()).view(T, B, 4 * H).contiguous()
+        h = h0.contiguous().clone()
+        c = c0.contiguous().clone()
+        hs = torch.empty(T, B, H, device=dev)
+        hs_in = torch.empty(T, B, H, device=dev)
+        cs_in = torch.empty(T, B, H, device=dev)
+        cs_out = torch.empty(T, B, H, device=dev)
+        gemm_tmp = torch.empty(B, 4 * H, device=dev)
+        nd = notdone.reshape(T, B, 1)[:, :, 0].contiguous()
+        w_hh_c = w_hh.contiguous()
+        ret = lib.masked_lstm_seq_fwd(
+            _cp(xg), _cp(w_hh_c), _cp(nd), _cp(h), _cp(c), _cp(hs),
+            _cp(hs_in), _cp(cs_in), _cp(cs_out), _cp(gemm_tmp), T, B, H,
+            _backend.current_stream())
+        _backend.check(ret, "masked_lstm_seq_fwd")
+        ctx.save_for_backward(x, nd, hs_in, cs_in, cs_out, xg, w_ih, w_hh_c)
+        ctx.H = H
+        return hs, h, c
+
+    @staticmethod
+    @torch.amp.custom_bwd(device_type="cuda")
+    def backward(ctx, d_hs, d_hT, d_cT):
+        lib = _backend.lib()
+        x, nd, hs_in, cs_in, cs_out, gates_all, w_ih, w_hh = ctx.saved_tensors
+        T, B, I = x.shape
+        H = ctx.H
+        dev = x.device
+        dgates_all = torch.empty(T, B, 4 * H, device=dev)
+        dh_carry = d_hT.contiguous().clone()
+        dc_carry = d_cT.contiguous().clone()
+        dc_prev_tmp = torch.empty(B, H, device=dev)
+        ret = lib.masked_lstm_seq_bwd(
+            _cp(gates_all), _cp(cs_in), _cp(cs_out),
+            _cp(d_hs.contiguous()), _cp(nd), _cp(w_hh), _cp(dgates_all),
+            _cp(dh_carry), _cp(dc_carry), _cp(dc_prev_tmp), T, B, H,
+            _backend.current_stream())
+        _backend.check(ret, "masked_lstm_seq_bwd")
+        dg2 = dgates_all.reshape(T * B, 4 * H)
+        dx = (dg2 @ w_ih).view(T, B, I)
+        dw_ih = dg2.t() @ x.reshape(T * B, I)
+        dw_hh = dg2.t() @ hs_in.reshape(T * B, H)
+        db = dg2.sum(0)
+        return dx, None, dh_carry, dc_carry, dw_ih, dw_hh, db, db
+
+
+def _cp(t: torch.Tensor):
+    return _c(t.data_ptr())
+
+
+_c = ctypes.c_void_p
+_USE_SEQ = None
+
+
+def _use_seq_path() -> bool:
+    global _USE_SEQ
+    if _USE_SEQ is None:
+        import os
+        _USE_SEQ = bool(os.environ.get("SCALERL_LSTM_SEQ"))
+    return _USE_SEQ
+
+
 class MaskedLSTM(nn.Module):
     """Multi-layer done-masked LSTM with nn.LSTM-compatible parameters."""
 
@@ -170,8 +247,10 @@ class MaskedLSTM(nn.Module):
         hs_out: List[torch.Tensor] = []
         cs_out: List[torch.Tensor] = []
         out = x
+        fn = (_MaskedLSTMSeqFn if (x.is_cuda and _use_seq_path())
+              else _MaskedLSTMFn)
         for k in range(self.num_layers):
-            out, hN, cN = _MaskedLSTMFn.apply(
+            out, hN, cN = fn.apply(
                 out.contiguous(), nd, h0[k], c0[k],
                 getattr(self, f"weight_ih_l{k}"),
                 getattr(self, f"weight_hh_l{k}"),

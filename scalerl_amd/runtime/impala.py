@@ -13,8 +13,9 @@ design:
   → backward → flat-grad clip (HIP) → RCCL all-reduce → fused RMSProp (HIP);
 - weight publication: one D2H flat copy into the shared CPU flat buffer the
   actor models alias (impala_atari.py:348 equivalent);
-- process discipline: actors are forked BEFORE any HIP/CUDA init (fork
-  after GPU init is unsafe on ROCm, and actors never touch the GPU).
+- process discipline: children are SPAWNED whenever the GPU is involved
+  (the HIP runtime does not survive fork on ROCm); pure-CPU runs keep
+  fork for startup speed.
 
 Step accounting matches the reference (impala_atari.py:391): one learn
 iteration consumes rollout_length × batch_size env steps.

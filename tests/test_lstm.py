@@ -85,3 +85,36 @@ def test_masked_lstm_gpu_matches_cpu():
     torch.testing.assert_close(xg.grad.cpu(), xc.grad, rtol=1e-3, atol=1e-4)
     torch.testing.assert_close(mg.weight_hh_l0.grad.cpu(),
                                ml.weight_hh_l0.grad, rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not __import__("os").environ.get("SCALERL_EXPERIMENTAL"),
+                    reason="C++-loop LSTM path pending hardware validation")
+def test_masked_lstm_seq_path_matches_python_path():
+    """SCALERL_LSTM_SEQ C++-driven loop vs the per-step Python path."""
+    import scalerl_amd.ops.lstm as lstm_mod
+    torch.manual_seed(0)
+    T, B, I, H, L = 20, 8, 32, 64, 2
+    ml = MaskedLSTM(I, H, num_layers=L).cuda()
+    x = torch.randn(T, B, I, device="cuda")
+    notdone = (torch.rand(T, B, device="cuda") > 0.1).float()
+    state = tuple(s.cuda() for s in ml.initial_state(B))
+
+    def run():
+        ml.zero_grad()
+        xg = x.clone().requires_grad_()
+        out, (h, c) = ml(xg, notdone, state)
+        (out.square().mean() + h.sum() * 0.1).backward()
+        return (out.detach().cpu(), h.detach().cpu(),
+                xg.grad.cpu(), ml.weight_hh_l0.grad.cpu().clone(),
+                ml.weight_ih_l1.grad.cpu().clone())
+
+    lstm_mod._USE_SEQ = False
+    ref = run()
+    lstm_mod._USE_SEQ = True
+    try:
+        got = run()
+    finally:
+        lstm_mod._USE_SEQ = None
+    for g, r in zip(got, ref):
+        torch.testing.assert_close(g, r, rtol=1e-4, atol=1e-5)
