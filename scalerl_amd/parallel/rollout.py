@@ -11,8 +11,9 @@ MI355X redesign of the reference's buffer machinery
 - the learner hipHostRegisters the shared region so H2D copies are true
   async DMA on a dedicated side stream (replaces `.to(device,
   non_blocking=True)` at impala_atari.py:259-266);
-- free/full index queues carry slot ids (SimpleQueue, as the reference) —
-  latency is amortized over T*E env steps per slot;
+- free/full index queues carry slot ids (free: SimpleQueue; full: a
+  timeout-capable Queue the learner watchdogs) — latency is amortized over
+  T*E env steps per slot;
 - weight publication: actors' model params alias ONE shared flat fp32
   buffer (parallel/flat.py); the learner publishes with a single flat copy
   (replaces load_state_dict at impala_atari.py:348).
