@@ -157,6 +157,11 @@ class ImpalaArguments(RLArguments):
     output_dir: str = _h("checkpoint/log dir", default="work_dirs/impala")
     checkpoint_path: str = _h("explicit checkpoint file (empty → auto)", default="")
     inference: str = _h("actor inference placement: cpu | gpu", default="cpu")
+    remote_actor_slots: int = _h("store slots reserved for remote-node "
+                                 "actors (0 disables the TCP server)", default=0)
+    remote_port: int = _h("TCP port for remote actor nodes (0 = ephemeral)", default=0)
+    remote_publish_interval: int = _h("learn steps between TCP weight "
+                                      "publications", default=10)
     dtype: str = _h("learner compute dtype: bf16 | fp32", default="bf16")
     use_graph: bool = _h("hipGraph-capture the learner step "
                          "(experimental: conflicts with the GPU inference "

@@ -114,10 +114,12 @@ class WorkerServer:
     (worker.py:269-297 structure, one port, typed frames)."""
 
     def __init__(self, config: Dict[str, Any], port: int = 9999,
-                 episode_callback: Optional[Callable] = None):
+                 episode_callback: Optional[Callable] = None,
+                 retain_episodes: bool = True):
         self.config = config
         self.param_server = ParameterServer()
         self.episode_callback = episode_callback
+        self.retain_episodes = retain_episodes
         self.episodes: List[Tuple[Dict, List[torch.Tensor]]] = []
         self._next_worker_id = 0
         self._lock = threading.Lock()
@@ -159,8 +161,9 @@ class WorkerServer:
                     conn.send({"kind": "weights", "version": v},
                               [w] if w is not None else [])
                 elif kind == "episode":
-                    with self._lock:
-                        self.episodes.append((header, tensors))
+                    if self.retain_episodes:
+                        with self._lock:
+                            self.episodes.append((header, tensors))
                     if self.episode_callback is not None:
                         self.episode_callback(header, tensors)
                     conn.send({"kind": "episode_ack"})

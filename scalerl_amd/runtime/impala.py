@@ -75,6 +75,8 @@ class ImpalaTrainer:
         # (8 ranks x 24 actors x 290 MB slots must fit /dev/shm)
         num_buffers = args.num_buffers or (
             args.num_actors + 2 * self.slots_per_batch + 2)
+        self.local_slots = num_buffers
+        num_buffers += args.remote_actor_slots  # reserved tail ids
 
         # ---- CPU phase: shared actor model + store + actor processes ----
         # (everything here must precede any CUDA/HIP initialization)
@@ -170,7 +172,7 @@ class ImpalaTrainer:
                 name=f"impala-actor-{self.rank}-{i}")
             p.start()
             self.actors.append(p)
-        for s in range(self._num_buffers):
+        for s in range(self.local_slots):
             self.free_q.put(s)
         self._started = True
 
@@ -201,6 +203,28 @@ class ImpalaTrainer:
         self._publish_weights()
         self.autocast_dtype = (torch.bfloat16 if args.dtype == "bf16" and
                                self.device.type == "cuda" else None)
+
+        # optional TCP server for remote-node actor farms
+        self.remote_server = None
+        if args.remote_actor_slots > 0:
+            import queue as _q
+            from ..parallel.remote_actors import RemoteSlotServer
+            self._free_remote_q = _q.Queue()
+            for s in range(self.local_slots, self._num_buffers):
+                self._free_remote_q.put(s)
+            lstm_hidden = self.store.lstm_hidden
+            config = {"env_id": args.env_id,
+                      "rollout_length": args.rollout_length,
+                      "envs_per_actor": args.envs_per_actor,
+                      "obs_shape": list(self.obs_shape),
+                      "num_actions": self.num_actions,
+                      "lstm_hidden": lstm_hidden, "seed": args.seed}
+            self.remote_server = RemoteSlotServer(
+                self.store, self.full_q, self._free_remote_q,
+                self.shared_flat.flat, config, port=args.remote_port)
+            self.remote_server.publish_weights()
+            self.log.info(f"remote actor server on port "
+                          f"{self.remote_server.port}")
 
     def _pause_inference(self, pause: bool, timeout_s: float = 30.0) -> None:
         """Quiesce the inference worker's HIP queue around graph capture."""
@@ -269,7 +293,10 @@ class ImpalaTrainer:
         batch = self.gatherer.finish(token)
         self.timings.time("gather_wait")
         for s in slot_ids:
-            self.free_q.put(s)
+            if s >= self.local_slots:  # reserved remote-upload slot
+                self._free_remote_q.put(s)
+            else:
+                self.free_q.put(s)
         self._pending = self._start_prefetch()
         return batch
 
@@ -344,6 +371,9 @@ class ImpalaTrainer:
         self.optimizer.step(self.flat.flat_grad, lr=lr)
         self.timings.time("optimize")
         self._publish_weights()
+        if (self.remote_server is not None and
+                self.learn_iters % max(args.remote_publish_interval, 1) == 0):
+            self.remote_server.publish_weights()
         self.timings.time("publish")
 
         self.gatherer.mark_consumed()
@@ -485,6 +515,9 @@ class ImpalaTrainer:
             if self.inference_proc.is_alive():
                 self.inference_proc.terminate()
             self.inference_proc = None
+        if getattr(self, "remote_server", None) is not None:
+            self.remote_server.close()
+            self.remote_server = None
         if getattr(self, "_pins", None) is not None:
             if self.device.type == "cuda":
                 torch.cuda.synchronize()

@@ -60,3 +60,38 @@ def test_frame_connection_large_tensor():
         c.close()
     finally:
         srv.close()
+
+
+def test_remote_actor_node_feeds_learner(tmp_path):
+    """Multi-node IMPALA over localhost: a remote actor node ships rollout
+    slots into the learner's store; the learner consumes them through the
+    same full_q path as local slots and recycles the reserved ids."""
+    from scalerl_amd.config import ImpalaArguments
+    from scalerl_amd.parallel.remote_actors import remote_actor_node
+    from scalerl_amd.runtime.impala import ImpalaTrainer
+
+    args = ImpalaArguments(rollout_length=8, batch_size=8, envs_per_actor=8,
+                           num_actors=1, total_steps=1 << 40, use_lstm=True,
+                           device="cpu", dtype="fp32",
+                           output_dir=str(tmp_path), seed=5,
+                           checkpoint_interval_s=1e9,
+                           remote_actor_slots=2, remote_port=0)
+    t = ImpalaTrainer(args)
+    try:
+        t.start_actors()
+        t.setup_learner()
+        port = t.remote_server.port
+
+        # ship exactly as many slots as are reserved (further uploads
+        # would block on backpressure until the learner recycles)
+        shipped = remote_actor_node("127.0.0.1", port, num_actors=1,
+                                    max_slots=2)
+        assert shipped == 2
+        for _ in range(6):  # consume local + remote slots
+            t.train_iteration()
+        assert t.global_step == 6 * 8 * 8
+        # remote ids recycled to the server queue (one may still be held
+        # by the in-flight prefetch)
+        assert t._free_remote_q.qsize() >= args.remote_actor_slots - 1
+    finally:
+        t.shutdown()
