@@ -148,3 +148,20 @@ def test_impala_single_actor_reproducible(tmp_path):
     a, b, c = run(11), run(11), run(12)
     assert a == b
     assert a != c
+
+
+def test_impala_train_driver_loop(tmp_path):
+    """The full train() driver: loop to total_steps, periodic + final
+    checkpoint, clean shutdown."""
+    args = _args(tmp_path, total_steps=8 * 8 * 3, checkpoint_interval_s=0.0,
+                 disable_checkpoint=False)
+    args.output_dir = str(tmp_path)
+    t = ImpalaTrainer(args)
+    t.train()  # runs start_actors + setup_learner + loop + shutdown
+    assert t.global_step >= args.total_steps
+    ckpt = os.path.join(str(tmp_path), "model.tar")
+    assert os.path.exists(ckpt)
+    loaded = torch.load(ckpt, map_location="cpu", weights_only=False)
+    assert loaded["hparam"]["rollout_length"] == 8
+    assert loaded["global_step"] == t.global_step
+    assert not t.actors  # shutdown joined everything
