@@ -118,24 +118,32 @@ class AtariNet(nn.Module):
 
 
 class AtariQNet(nn.Module):
-    """Nature-CNN Q-network for Ape-X (apex/network.py parity, dueling
-    option): uint8 [N,C,84,84] → Q [N,A]."""
+    """Nature-CNN Q-network for Ape-X (apex/network.py parity): uint8
+    [N,C,84,84] → Q [N,A], with dueling and NoisyNet head options
+    (Rainbow-style exploration for the image agent)."""
 
     def __init__(self, observation_shape=(4, 84, 84), num_actions: int = 6,
-                 dueling: bool = True):
+                 dueling: bool = True, noisy: bool = False):
         super().__init__()
+        from .noisy import NoisyLinear
         c = observation_shape[0]
         self.num_actions = num_actions
         self.dueling = dueling
+        self.noisy = noisy
+        lin = NoisyLinear if noisy else nn.Linear
         self.conv1 = nn.Conv2d(c, 32, kernel_size=8, stride=4)
         self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
         self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
         self.fc = nn.Linear(64 * 7 * 7, 512)
         if dueling:
-            self.value_head = nn.Linear(512, 1)
-            self.adv_head = nn.Linear(512, num_actions)
+            self.value_head = lin(512, 1)
+            self.adv_head = lin(512, num_actions)
         else:
-            self.head = nn.Linear(512, num_actions)
+            self.head = lin(512, num_actions)
+
+    def reset_noise(self):
+        from .noisy import reset_noise
+        reset_noise(self)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = x.float() / 255.0

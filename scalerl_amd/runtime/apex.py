@@ -53,6 +53,12 @@ class EpsGreedyPolicy:
 
     @torch.no_grad()
     def __call__(self, obs, reward, done, last_action, want_state=False):
+        if getattr(self.model, "noisy", False):
+            # NoisyNet exploration: resample parameter noise instead of ε
+            self.model.train()
+            self.model.reset_noise()
+            q = self.model(obs)
+            return q.argmax(dim=-1), q, None
         q = self.model(obs)
         action = q.argmax(dim=-1)
         explore = torch.from_numpy(
@@ -104,7 +110,8 @@ class ApexTrainer:
         self.chunk_len = 32  # transitions per slot row-block
         torch.manual_seed(args.seed + self.rank)
         self.actor_model = AtariQNet(self.obs_shape, self.num_actions,
-                                     dueling=args.dueling_dqn)
+                                     dueling=args.dueling_dqn,
+                                     noisy=args.noisy_dqn)
         self.actor_model.eval()
         self.shared_flat = FlatParams(self.actor_model, device="cpu",
                                       share=True)
@@ -151,10 +158,12 @@ class ApexTrainer:
         if dev.type == "cuda":
             torch.backends.cudnn.benchmark = True
         self.model = AtariQNet(self.obs_shape, self.num_actions,
-                               dueling=args.dueling_dqn).to(dev)
+                               dueling=args.dueling_dqn,
+                               noisy=args.noisy_dqn).to(dev)
         self.model.load_state_dict(self.actor_model.state_dict())
         self.target_model = AtariQNet(self.obs_shape, self.num_actions,
-                                      dueling=args.dueling_dqn).to(dev)
+                                      dueling=args.dueling_dqn,
+                                      noisy=args.noisy_dqn).to(dev)
         self.target_model.load_state_dict(self.model.state_dict())
         self.flat = FlatParams(self.model, device=dev)
         self.target_flat = FlatParams(self.target_model, device=dev)
