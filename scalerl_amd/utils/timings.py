@@ -66,6 +66,40 @@ class Timer:
         return False
 
 
+_TIMER_REGISTRY = {}
+
+
+def check_time(name: str) -> "Timer":
+    """Named global timers (parity with utils/timer.py's check_time
+    registry): ``with check_time("phase"): ...`` accumulates per name."""
+    t = _TIMER_REGISTRY.get(name)
+    if t is None:
+        t = _AccumTimer(name)
+        _TIMER_REGISTRY[name] = t
+    return t
+
+
+def timer_report() -> str:
+    return "\n".join(f"{n}: {t.total:.4f}s over {t.count} calls"
+                      for n, t in sorted(_TIMER_REGISTRY.items()))
+
+
+class _AccumTimer:
+    def __init__(self, name: str):
+        self.name = name
+        self.total = 0.0
+        self.count = 0
+
+    def __enter__(self):
+        self._t0 = time.perf_counter()
+        return self
+
+    def __exit__(self, *exc):
+        self.total += time.perf_counter() - self._t0
+        self.count += 1
+        return False
+
+
 class CudaTimings:
     """hipEvent-based per-section device timings (lazy sync).
 
