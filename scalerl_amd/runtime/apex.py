@@ -118,8 +118,11 @@ class ApexTrainer:
         self.store = RolloutStore(
             2 * args.num_actors + 4, self.chunk_len, E, self.obs_shape,
             self.num_actions, lstm_hidden=0)
+        import os as _os
         self._mp_ctx = ("fork" if (self.device.type == "cpu"
-                                   and not torch.cuda.is_initialized())
+                                   and not torch.cuda.is_initialized()
+                                   and not _os.environ.get(
+                                       "SCALERL_FORCE_SPAWN"))
                         else "spawn")
         ctx = mp.get_context(self._mp_ctx)
         self.free_q = ctx.SimpleQueue()

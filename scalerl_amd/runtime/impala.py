@@ -96,7 +96,9 @@ class ImpalaTrainer:
         # survive fork on ROCm.  Shared-memory tensors travel to spawned
         # children as shm handles.  Pure-CPU runs keep fork (fast startup).
         self._mp_ctx = ("fork" if (self.device.type == "cpu"
-                                   and not torch.cuda.is_initialized())
+                                   and not torch.cuda.is_initialized()
+                                   and not os.environ.get(
+                                       "SCALERL_FORCE_SPAWN"))
                         else "spawn")
         ctx = mp.get_context(self._mp_ctx)
         self.free_q = ctx.SimpleQueue()

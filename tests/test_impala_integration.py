@@ -165,3 +165,18 @@ def test_impala_train_driver_loop(tmp_path):
     assert loaded["hparam"]["rollout_length"] == 8
     assert loaded["global_step"] == t.global_step
     assert not t.actors  # shutdown joined everything
+
+
+def test_impala_spawn_context_picklable(tmp_path, monkeypatch):
+    """Everything handed to spawned children must pickle (the GPU-box
+    topology); guard the exact regression that fork-only testing missed."""
+    monkeypatch.setenv("SCALERL_FORCE_SPAWN", "1")
+    t = ImpalaTrainer(_args(tmp_path))
+    try:
+        assert t._mp_ctx == "spawn"
+        t.start_actors()
+        t.setup_learner()
+        stats = t.train_iteration()
+        assert torch.isfinite(stats["total_loss"])
+    finally:
+        t.shutdown()
