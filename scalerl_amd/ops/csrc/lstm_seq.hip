@@ -204,8 +204,9 @@ extern "C" int masked_lstm_seq_bwd(
     // the fresh dc_prev, which becomes the next (earlier) step's dc carry
     hipLaunchKernelGGL(lstm_mask_carry_kernel, dim3(grid), dim3(block), 0,
                        stream, dh_carry, dc_prev_tmp, notdone + t * B, B, H);
-    hipMemcpyAsync(dc_carry, dc_prev_tmp, B * H * sizeof(float),
-                   hipMemcpyDeviceToDevice, stream);
+    if (hipMemcpyAsync(dc_carry, dc_prev_tmp, B * H * sizeof(float),
+                       hipMemcpyDeviceToDevice, stream) != hipSuccess)
+      return -13;
   }
   CHECK_LAUNCH();
   return 0;

@@ -27,13 +27,17 @@ from . import _backend
 _c = ctypes.c_void_p
 
 
-def _pointwise_fwd(gates: torch.Tensor, c_prev: torch.Tensor):
-    """Returns (h, c); on GPU `gates` is overwritten with activated gates."""
+def _pointwise_fwd(gates: torch.Tensor, c_prev: torch.Tensor,
+                   out_h: torch.Tensor = None, out_c: torch.Tensor = None):
+    """Returns (h, c); `gates` is overwritten with activated gates.  When
+    ``out_h/out_c`` are given the results are written there (one fewer
+    launch per step on the hot path)."""
     B, H4 = gates.shape
     H = H4 // 4
     if gates.is_cuda:
-        h = torch.empty(B, H, device=gates.device, dtype=torch.float32)
-        c = torch.empty_like(h)
+        h = out_h if out_h is not None else torch.empty(
+            B, H, device=gates.device, dtype=torch.float32)
+        c = out_c if out_c is not None else torch.empty_like(h)
         ret = _backend.lib().lstm_pointwise_fwd(
             _c(gates.data_ptr()), _c(c_prev.data_ptr()), _c(h.data_ptr()),
             _c(c.data_ptr()), B, H, _backend.current_stream())
@@ -41,18 +45,28 @@ def _pointwise_fwd(gates: torch.Tensor, c_prev: torch.Tensor):
         return h, c
     i, f, g, o = gates.chunk(4, dim=1)
     i, f, g, o = i.sigmoid(), f.sigmoid(), g.tanh(), o.sigmoid()
-    c = f * c_prev + i * g
-    h = o * torch.tanh(c)
+    c_new = f * c_prev + i * g
+    h_new = o * torch.tanh(c_new)
     gates.copy_(torch.cat([i, f, g, o], dim=1))  # match GPU in-place contract
-    return h, c
+    if out_h is not None:
+        out_h.copy_(h_new)
+        h_new = out_h
+    if out_c is not None:
+        out_c.copy_(c_new)
+        c_new = out_c
+    return h_new, c_new
 
 
-def _pointwise_bwd(gates_act, c_prev, c_out, dh, dc_in):
+def _pointwise_bwd(gates_act, c_prev, c_out, dh, dc_in,
+                   out_dgates: torch.Tensor = None,
+                   out_dc_prev: torch.Tensor = None):
     B, H4 = gates_act.shape
     H = H4 // 4
     if gates_act.is_cuda:
-        dgates = torch.empty_like(gates_act)
-        dc_prev = torch.empty(B, H, device=gates_act.device, dtype=torch.float32)
+        dgates = out_dgates if out_dgates is not None else \
+            torch.empty_like(gates_act)
+        dc_prev = out_dc_prev if out_dc_prev is not None else torch.empty(
+            B, H, device=gates_act.device, dtype=torch.float32)
         ret = _backend.lib().lstm_pointwise_bwd(
             _c(gates_act.data_ptr()), _c(c_prev.data_ptr()), _c(c_out.data_ptr()),
             _c(dh.data_ptr()), _c(dc_in.data_ptr()) if dc_in is not None else None,
@@ -69,7 +83,15 @@ def _pointwise_bwd(gates_act, c_prev, c_out, dh, dc_in):
     df = dc * c_prev * f * (1 - f)
     dg = dc * i * (1 - g * g)
     do = dh * tc * o * (1 - o)
-    return torch.cat([di, df, dg, do], dim=1), dc * f
+    dgates = torch.cat([di, df, dg, do], dim=1)
+    dc_prev = dc * f
+    if out_dgates is not None:
+        out_dgates.copy_(dgates)
+        dgates = out_dgates
+    if out_dc_prev is not None:
+        out_dc_prev.copy_(dc_prev)
+        dc_prev = out_dc_prev
+    return dgates, dc_prev
 
 
 class _MaskedLSTMFn(torch.autograd.Function):
@@ -92,21 +114,22 @@ class _MaskedLSTMFn(torch.autograd.Function):
         cs_out = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
         hs_in = torch.empty(T, B, H, device=x.device, dtype=x.dtype)
         gates_all = torch.empty(T, B, 4 * H, device=x.device, dtype=x.dtype)
+        w_hh_t = w_hh.t()
         for t in range(T):
             nd = notdone[t]
-            h = h * nd
-            c = c * nd
-            hs_in[t] = h
-            cs_in[t] = c
-            gates = torch.addmm(xg[t], h, w_hh.t())
-            gates_all[t] = gates
-            h, c = _pointwise_fwd(gates_all[t], cs_in[t])
-            hs[t] = h
-            cs_out[t] = c
+            # out=-form writes land directly in the saved [T,...] buffers:
+            # 4 enqueues per step instead of 8 (the unroll is launch-bound)
+            h = torch.mul(h, nd, out=hs_in[t])
+            c = torch.mul(c, nd, out=cs_in[t])
+            torch.addmm(xg[t], h, w_hh_t, out=gates_all[t])
+            h, c = _pointwise_fwd(gates_all[t], c, out_h=hs[t],
+                                  out_c=cs_out[t])
         ctx.save_for_backward(x, notdone, hs_in, cs_in, cs_out, gates_all,
                               w_ih, w_hh)
         ctx.H = H
-        return hs, h, c
+        # final h/c alias hs[T-1]/cs_out[T-1] after the out=-form loop;
+        # clone so the Function's outputs don't share storage
+        return hs, h.clone(), c.clone()
 
     @staticmethod
     @torch.amp.custom_bwd(device_type="cuda")
@@ -115,17 +138,20 @@ class _MaskedLSTMFn(torch.autograd.Function):
          w_hh) = ctx.saved_tensors
         T, B, I = x.shape
         H = ctx.H
-        dh_carry = d_hT.contiguous()
-        dc_carry = d_cT.contiguous()
+        dh_carry = d_hT.contiguous().clone()
+        dc_carry = d_cT.contiguous().clone()
         dgates_all = torch.empty_like(gates_all)
+        dh_buf = torch.empty_like(dh_carry)
+        dc_prev_buf = torch.empty_like(dc_carry)
         for t in range(T - 1, -1, -1):
-            dh = d_hs[t] + dh_carry
-            dgates, dc_prev = _pointwise_bwd(gates_all[t], cs_in[t], cs_out[t],
-                                             dh.contiguous(), dc_carry)
-            dgates_all[t] = dgates
+            dh = torch.add(d_hs[t], dh_carry, out=dh_buf)
+            dgates, dc_prev = _pointwise_bwd(
+                gates_all[t], cs_in[t], cs_out[t], dh, dc_carry,
+                out_dgates=dgates_all[t], out_dc_prev=dc_prev_buf)
             nd = notdone[t]
-            dh_carry = (dgates @ w_hh) * nd
-            dc_carry = dc_prev * nd
+            torch.mm(dgates, w_hh, out=dh_carry)
+            dh_carry.mul_(nd)
+            torch.mul(dc_prev, nd, out=dc_carry)
         dg2 = dgates_all.reshape(T * B, 4 * H)
         dx = (dg2 @ w_ih).view(T, B, I)
         dw_ih = dg2.t() @ x.reshape(T * B, I)
