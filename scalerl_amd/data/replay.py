@@ -200,15 +200,19 @@ class MultiStepReplayBuffer(ReplayBuffer):
             folded_d.reshape(-1), discount=disc.reshape(-1))
 
 
-class PrioritizedReplayBuffer(ReplayBuffer):
+class PrioritizedReplayBuffer(MultiStepReplayBuffer):
     """Proportional PER (replay_buffer.py:276-381 semantics) over the
     device sum tree.  Priorities stored as p^alpha; new transitions get
     max_priority^alpha; stratified sampling; IS weights computed in the
-    fused TD-loss kernel from (prio, total, min)."""
+    fused TD-loss kernel from (prio, total, min).
+
+    Inherits the n-step fold-at-insert front end (n_steps=1 → plain 1-step)
+    so PER and n-step compose, as in the reference where
+    PrioritizedReplayBuffer subclasses MultiStepReplayBuffer."""
 
     def __init__(self, capacity: int, obs_shape, alpha: float = 0.6,
-                 **kw):
-        super().__init__(capacity, obs_shape, **kw)
+                 n_steps: int = 1, **kw):
+        super().__init__(capacity, obs_shape, n_steps=n_steps, **kw)
         from ..ops import SumTree
         self.alpha = alpha
         self.tree = SumTree(capacity, device=self.device)

@@ -248,9 +248,12 @@ class MaskedLSTM(nn.Module):
             in_sz = input_size if k == 0 else hidden_size
             w_ih = nn.Parameter(torch.empty(4 * hidden_size, in_sz))
             w_hh = nn.Parameter(torch.empty(4 * hidden_size, hidden_size))
-            b_ih = nn.Parameter(torch.zeros(4 * hidden_size))
-            b_hh = nn.Parameter(torch.zeros(4 * hidden_size))
-            for w in (w_ih, w_hh):
+            b_ih = nn.Parameter(torch.empty(4 * hidden_size))
+            b_hh = nn.Parameter(torch.empty(4 * hidden_size))
+            # match nn.LSTM defaults: ALL params uniform(-H^-0.5, H^-0.5),
+            # biases included, so fresh-training dynamics match the
+            # reference's torch LSTM core (atari_model.py:51-55)
+            for w in (w_ih, w_hh, b_ih, b_hh):
                 nn.init.uniform_(w, -hidden_size ** -0.5, hidden_size ** -0.5)
             setattr(self, f"weight_ih_l{k}", w_ih)
             setattr(self, f"weight_hh_l{k}", w_hh)

@@ -22,7 +22,11 @@ actors→node, both as raw tensor frames.
 
 from __future__ import annotations
 
-import pickle
+import hashlib
+import hmac
+import json
+import os
+import secrets as _secrets
 import socket
 import struct
 import threading
@@ -31,21 +35,79 @@ from typing import Any, Callable, Dict, List, Optional, Tuple
 import torch
 
 _HDR = struct.Struct("!Q")
+# untrusted-input bounds: a peer can never make us allocate more than this
+_MAX_META = 16 << 20      # 16 MiB of JSON metadata
+_MAX_TENSOR = 4 << 30     # 4 GiB per tensor frame
+# dtype whitelist — `getattr(torch, name)` on attacker bytes is not ok
+_DTYPES = {
+    "float32": torch.float32, "float64": torch.float64,
+    "float16": torch.float16, "bfloat16": torch.bfloat16,
+    "int64": torch.int64, "int32": torch.int32, "int16": torch.int16,
+    "int8": torch.int8, "uint8": torch.uint8, "bool": torch.bool,
+}
+
+
+def cluster_secret() -> bytes:
+    """Shared secret for the HMAC handshake (SCALERL_CLUSTER_SECRET).
+    Every node of a multi-node run must export the same value."""
+    return os.environ.get("SCALERL_CLUSTER_SECRET", "").encode()
 
 
 class FrameConnection:
-    """Length-prefixed frames; payloads are (header dict, tensor list)."""
+    """Length-prefixed frames; payloads are (header dict, tensor list).
+
+    Wire safety (this is the only code that parses bytes off the network):
+    metadata frames are JSON — never pickle, so a malicious peer can at
+    worst send bad values, not code; tensor frames are raw buffers decoded
+    against a dtype whitelist with size caps.  Connections are authenticated
+    with an HMAC-SHA256 challenge/response over a shared secret before any
+    frame is parsed (reference's PickledConnection, connection.py:26-84,
+    had neither property).
+    """
 
     def __init__(self, sock: socket.socket):
         self.sock = sock
         self.lock = threading.Lock()
 
     @classmethod
-    def connect(cls, host: str, port: int, timeout: float = 30.0):
+    def connect(cls, host: str, port: int, timeout: float = 30.0,
+                secret: Optional[bytes] = None):
         s = socket.create_connection((host, port), timeout=timeout)
         s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
-        return cls(s)
+        conn = cls(s)
+        conn._client_auth(secret if secret is not None else cluster_secret())
+        return conn
 
+    # -- authentication ----------------------------------------------------
+    def _client_auth(self, secret: bytes) -> None:
+        nonce = self._recv_exact(32)
+        mac = hmac.new(secret, nonce, hashlib.sha256).digest()
+        self.sock.sendall(mac)
+        ok = self._recv_exact(1)
+        if ok != b"\x01":
+            raise ConnectionError(
+                "cluster auth rejected — set the same SCALERL_CLUSTER_SECRET "
+                "on every node")
+
+    def server_auth(self, secret: bytes) -> bool:
+        """Server side of the handshake; returns False on bad MAC."""
+        nonce = _secrets.token_bytes(32)
+        self.sock.sendall(nonce)
+        want = hmac.new(secret, nonce, hashlib.sha256).digest()
+        try:
+            got = self._recv_exact(32)
+        except (ConnectionResetError, OSError):
+            return False
+        if not hmac.compare_digest(want, got):
+            try:
+                self.sock.sendall(b"\x00")
+            except OSError:
+                pass
+            return False
+        self.sock.sendall(b"\x01")
+        return True
+
+    # -- framing -----------------------------------------------------------
     def _recv_exact(self, n: int) -> bytes:
         buf = bytearray()
         while len(buf) < n:
@@ -61,7 +123,7 @@ class FrameConnection:
         meta = {"h": header,
                 "t": [(list(t.shape), str(t.dtype).replace("torch.", ""))
                       for t in tensors]}
-        mb = pickle.dumps(meta, protocol=pickle.HIGHEST_PROTOCOL)
+        mb = json.dumps(meta).encode()
         with self.lock:
             self.sock.sendall(_HDR.pack(len(mb)) + mb)
             for t in tensors:
@@ -70,13 +132,19 @@ class FrameConnection:
 
     def recv(self) -> Tuple[Dict[str, Any], List[torch.Tensor]]:
         n = _HDR.unpack(self._recv_exact(_HDR.size))[0]
-        meta = pickle.loads(self._recv_exact(n))
+        if n > _MAX_META:
+            raise ConnectionError(f"metadata frame too large ({n} B)")
+        meta = json.loads(self._recv_exact(n).decode())
         tensors = []
         for shape, dtype in meta["t"]:
+            if dtype not in _DTYPES:
+                raise ConnectionError(f"disallowed tensor dtype {dtype!r}")
             nb = _HDR.unpack(self._recv_exact(_HDR.size))[0]
+            if nb > _MAX_TENSOR:
+                raise ConnectionError(f"tensor frame too large ({nb} B)")
             raw = self._recv_exact(nb)
             t = torch.frombuffer(bytearray(raw),
-                                 dtype=getattr(torch, dtype)).reshape(shape)
+                                 dtype=_DTYPES[dtype]).reshape(shape)
             tensors.append(t)
         return meta["h"], tensors
 
@@ -115,7 +183,8 @@ class WorkerServer:
 
     def __init__(self, config: Dict[str, Any], port: int = 9999,
                  episode_callback: Optional[Callable] = None,
-                 retain_episodes: bool = True):
+                 retain_episodes: bool = True, bind: Optional[str] = None,
+                 secret: Optional[bytes] = None):
         self.config = config
         self.param_server = ParameterServer()
         self.episode_callback = episode_callback
@@ -124,7 +193,11 @@ class WorkerServer:
         self._next_worker_id = 0
         self._lock = threading.Lock()
         self._stop = threading.Event()
-        self._srv = socket.create_server(("0.0.0.0", port), backlog=64)
+        self._secret = secret if secret is not None else cluster_secret()
+        # default bind is loopback; cross-node runs opt in with
+        # bind="0.0.0.0" (or SCALERL_CLUSTER_BIND) + a shared secret
+        bind = bind or os.environ.get("SCALERL_CLUSTER_BIND", "127.0.0.1")
+        self._srv = socket.create_server((bind, port), backlog=64)
         self._srv.settimeout(0.5)
         self.port = self._srv.getsockname()[1]
         self._threads: List[threading.Thread] = []
@@ -140,10 +213,16 @@ class WorkerServer:
                 continue
             except OSError:
                 break
-            t = threading.Thread(target=self._serve,
+            t = threading.Thread(target=self._auth_and_serve,
                                  args=(FrameConnection(sock),), daemon=True)
             t.start()
             self._threads.append(t)
+
+    def _auth_and_serve(self, conn: FrameConnection):
+        if not conn.server_auth(self._secret):
+            conn.close()
+            return
+        self._serve(conn)
 
     def _serve(self, conn: FrameConnection):
         try:

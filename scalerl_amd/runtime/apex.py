@@ -187,6 +187,7 @@ class ApexTrainer:
             self._pins = PinRegistry()
             self._pins.pin_store(self.store)
             self._pins.pin(self.shared_flat.flat)
+            self._ingest_done = torch.cuda.Event()
         self._publish()
 
     @torch.no_grad()
@@ -258,6 +259,13 @@ class ApexTrainer:
                     raise RuntimeError(f"actor process(es) died: {dead}")
                 continue
             ingested += self.ingest_slot(slot)
+            # the H2D copies out of the hipHostRegister'd shared slot are
+            # async: fence them before recycling the slot, or an actor can
+            # overwrite it mid-copy and corrupt replay data (the PER path
+            # was only safe by accident via float(prio.max()))
+            if self.device.type == "cuda":
+                self._ingest_done.record(torch.cuda.current_stream())
+                self._ingest_done.synchronize()
             self.free_q.put(slot)
             self.global_step += self.chunk_len * args.envs_per_actor
         stats: Dict[str, float] = {"ingested": ingested}

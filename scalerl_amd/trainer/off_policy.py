@@ -27,8 +27,11 @@ class OffPolicyTrainer(BaseTrainer):
         use_per = getattr(args, "use_per", False)
         n_steps = getattr(args, "n_steps", 1)
         if use_per:
+            # PER composes with n-step (reference: PrioritizedReplayBuffer
+            # subclasses MultiStepReplayBuffer)
             self.buffer = PrioritizedReplayBuffer(
                 args.buffer_size, obs_shape, alpha=args.per_alpha,
+                n_steps=n_steps, num_envs=args.num_envs,
                 device=device, gamma=args.gamma, seed=args.seed)
         elif n_steps > 1:
             self.buffer = MultiStepReplayBuffer(
@@ -46,7 +49,7 @@ class OffPolicyTrainer(BaseTrainer):
         self.grad_steps = 0
 
     def store_experience(self, obs, action, reward, next_obs, done) -> None:
-        if self.n_steps > 1 and not self.use_per:
+        if self.n_steps > 1:  # n-step folds per env (PER or uniform)
             for i in range(len(action)):
                 self.buffer.add(obs[i], action[i], float(reward[i]),
                                 next_obs[i], float(done[i]), env_id=i)

@@ -95,3 +95,45 @@ def test_remote_actor_node_feeds_learner(tmp_path):
         assert t._free_remote_q.qsize() >= args.remote_actor_slots - 1
     finally:
         t.shutdown()
+
+
+def test_cluster_auth_rejects_bad_secret(monkeypatch):
+    """HMAC handshake: a peer with the wrong shared secret is refused
+    before any frame is parsed."""
+    import pytest
+    from scalerl_amd.parallel.cluster import FrameConnection
+    srv = WorkerServer({}, port=0, secret=b"right-secret")
+    try:
+        with pytest.raises((ConnectionError, OSError)):
+            conn = FrameConnection.connect("127.0.0.1", srv.port,
+                                           secret=b"wrong-secret")
+            conn.send({"kind": "entry"})
+            conn.recv()
+        # correct secret still works
+        c2 = FrameConnection.connect("127.0.0.1", srv.port,
+                                     secret=b"right-secret")
+        c2.send({"kind": "entry"})
+        ack, _ = c2.recv()
+        assert ack["kind"] == "entry_ack"
+        c2.close()
+    finally:
+        srv.close()
+
+
+def test_cluster_rejects_non_whitelisted_dtype():
+    """Wire decoding never getattr()s arbitrary names off attacker bytes."""
+    import json
+    import socket
+    import struct
+    import pytest
+    from scalerl_amd.parallel.cluster import FrameConnection
+    a, b = socket.socketpair()
+    try:
+        meta = json.dumps({"h": {}, "t": [[[1], "cuda"]]}).encode()
+        a.sendall(struct.Struct("!Q").pack(len(meta)) + meta)
+        conn = FrameConnection(b)
+        with pytest.raises(ConnectionError):
+            conn.recv()
+    finally:
+        a.close()
+        b.close()

@@ -95,3 +95,22 @@ def test_per_buffer_roundtrip_and_state():
     q.load_state_dict(sd)
     assert float(q.tree.total) == pytest.approx(float(p.tree.total))
     assert len(q) == len(p)
+
+
+def test_per_with_nstep_composes():
+    """PER + n_steps>1 fold together (reference: PrioritizedReplayBuffer
+    subclasses MultiStepReplayBuffer) — ADVICE r1 item 3."""
+    n, gamma = 3, 0.9
+    p = PrioritizedReplayBuffer(64, (1,), alpha=0.6, n_steps=n,
+                                gamma=gamma, seed=0)
+    rewards = [1.0, 2.0, 4.0, 8.0, 16.0]
+    for i, r in enumerate(rewards):
+        p.add([float(i)], 0, r, [float(i + 1)], 0.0)
+    # first full window: r0 + g*r1 + g^2*r2
+    want = rewards[0] + gamma * rewards[1] + gamma ** 2 * rewards[2]
+    assert float(p.reward[0]) == pytest.approx(want)
+    assert float(p.discount[0]) == pytest.approx(gamma ** n)
+    assert float(p.next_obs[0, 0]) == 3.0  # obs n steps ahead
+    # priorities assigned for the folded insert
+    batch, idx, prio, total, pmin = p.sample_with_priorities(2)
+    assert (prio > 0).all()
