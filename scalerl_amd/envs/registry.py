@@ -70,3 +70,22 @@ def make_env(env_id: str, seed: Optional[int] = None,
     if seed is not None:
         env.action_space.seed(seed)
     return env
+
+
+def make_gym_env(env_id: str, seed: int = 42, capture_video: bool = False,
+                 save_video_dir: str = "work_dir",
+                 save_video_name: str = "test",
+                 deepmind_wrap: bool = False):
+    """Reference ``make_gym_env`` parity (scalerl/envs/gym_env.py:6-33):
+    optional video capture + episode-statistics wrapper + action-space
+    seeding.  Video is written as animated GIFs (no encoder in the image);
+    the directory layout ``{save_video_dir}/{save_video_name}`` matches."""
+    from .recording import RecordEpisodeStatistics, RecordVideo
+    env = make_env(env_id, seed=seed, deepmind_wrap=deepmind_wrap)
+    if capture_video:
+        import os
+        env = RecordVideo(env, os.path.join(save_video_dir, save_video_name),
+                          name_prefix=save_video_name)
+    env = RecordEpisodeStatistics(env)
+    env.action_space.seed(seed)
+    return env

@@ -156,3 +156,48 @@ class AtariQNet(nn.Module):
             a = self.adv_head(h)
             return v + a - a.mean(dim=-1, keepdim=True)
         return self.head(h)
+
+
+class CategoricalAtariQNet(nn.Module):
+    """C51 distributional Q-network on the Nature-CNN body (the reference
+    declares v_min/v_max/atoms for the Atari-scale DQN, rl_args.py:221-260,
+    but ships no image implementation).  Same ``dist``/``forward``/support
+    surface as :class:`scalerl_amd.models.noisy.CategoricalQNet`, so
+    ``c51_loss`` works against either."""
+
+    def __init__(self, observation_shape=(4, 84, 84), num_actions: int = 6,
+                 num_atoms: int = 51, v_min: float = -10.0,
+                 v_max: float = 10.0, noisy: bool = False):
+        super().__init__()
+        from .noisy import NoisyLinear
+        c = observation_shape[0]
+        self.action_dim = self.num_actions = num_actions
+        self.num_atoms = num_atoms
+        self.noisy = noisy
+        lin = NoisyLinear if noisy else nn.Linear
+        self.conv1 = nn.Conv2d(c, 32, kernel_size=8, stride=4)
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
+        self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
+        self.fc = nn.Linear(64 * 7 * 7, 512)
+        self.head = lin(512, num_actions * num_atoms)
+        self.register_buffer("support", torch.linspace(v_min, v_max, num_atoms))
+        self.v_min, self.v_max = v_min, v_max
+        self.delta_z = (v_max - v_min) / (num_atoms - 1)
+
+    def reset_noise(self):
+        from .noisy import reset_noise
+        reset_noise(self)
+
+    def dist(self, x: torch.Tensor) -> torch.Tensor:
+        """→ log-probabilities [B, A, atoms]."""
+        x = x.float() / 255.0
+        x = F.relu(self.conv1(x))
+        x = F.relu(self.conv2(x))
+        x = F.relu(self.conv3(x))
+        h = F.relu(self.fc(torch.flatten(x, 1)))
+        logits = self.head(h).view(-1, self.num_actions, self.num_atoms)
+        return F.log_softmax(logits, dim=-1)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """Expected Q-values [B, A]."""
+        return (self.dist(x).exp() * self.support).sum(-1)

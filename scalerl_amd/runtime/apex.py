@@ -108,10 +108,9 @@ class ApexTrainer:
 
         E = args.envs_per_actor
         self.chunk_len = 32  # transitions per slot row-block
+        self.categorical = getattr(args, "categorical_dqn", False)
         torch.manual_seed(args.seed + self.rank)
-        self.actor_model = AtariQNet(self.obs_shape, self.num_actions,
-                                     dueling=args.dueling_dqn,
-                                     noisy=args.noisy_dqn)
+        self.actor_model = self._build_qnet()
         self.actor_model.eval()
         self.shared_flat = FlatParams(self.actor_model, device="cpu",
                                       share=True)
@@ -132,6 +131,16 @@ class ApexTrainer:
         self.actors: List[mp.Process] = []
         self.global_step = 0
         self.learn_iters = 0
+
+    def _build_qnet(self):
+        args = self.args
+        if self.categorical:
+            from ..models.atari import CategoricalAtariQNet
+            return CategoricalAtariQNet(
+                self.obs_shape, self.num_actions, num_atoms=args.num_atoms,
+                v_min=args.v_min, v_max=args.v_max, noisy=args.noisy_dqn)
+        return AtariQNet(self.obs_shape, self.num_actions,
+                         dueling=args.dueling_dqn, noisy=args.noisy_dqn)
 
     def actor_eps(self, i: int) -> float:
         n = max(self.args.num_actors - 1, 1)
@@ -160,13 +169,9 @@ class ApexTrainer:
         dev = self.device
         if dev.type == "cuda":
             torch.backends.cudnn.benchmark = True
-        self.model = AtariQNet(self.obs_shape, self.num_actions,
-                               dueling=args.dueling_dqn,
-                               noisy=args.noisy_dqn).to(dev)
+        self.model = self._build_qnet().to(dev)
         self.model.load_state_dict(self.actor_model.state_dict())
-        self.target_model = AtariQNet(self.obs_shape, self.num_actions,
-                                      dueling=args.dueling_dqn,
-                                      noisy=args.noisy_dqn).to(dev)
+        self.target_model = self._build_qnet().to(dev)
         self.target_model.load_state_dict(self.model.state_dict())
         self.flat = FlatParams(self.model, device=dev)
         self.target_flat = FlatParams(self.target_model, device=dev)
@@ -285,13 +290,31 @@ class ApexTrainer:
                 batch = self.buffer.sample(args.batch_size)
                 per_kw = {}
             self.flat.flat_grad.zero_()
-            q = self.model(batch["obs"])
-            with torch.no_grad():
-                qn_t = self.target_model(batch["next_obs"])
-                qn_o = self.model(batch["next_obs"]) if args.double_dqn else None
-            loss, td_abs = fused_td_loss(
-                q, qn_o, qn_t, batch["action"], batch["reward"],
-                batch["discount"], **per_kw)
+            if self.categorical:
+                # C51 projection loss (Bellemare et al. 2017); IS weights
+                # from the sampled priorities, priorities ← per-sample KL
+                from ..models.noisy import c51_loss
+                weights = None
+                if per_kw:
+                    N = per_kw["replay_size"]
+                    beta = per_kw["beta"]
+                    w = (per_kw["prios"] / per_kw["p_total"] * N) ** -beta
+                    w_max = (per_kw["p_min"] / per_kw["p_total"] * N) ** -beta
+                    weights = w / w_max
+                loss, td_abs = c51_loss(
+                    self.model, self.target_model, batch["obs"],
+                    batch["action"], batch["reward"], batch["discount"],
+                    batch["next_obs"], double=args.double_dqn,
+                    weights=weights)
+            else:
+                q = self.model(batch["obs"])
+                with torch.no_grad():
+                    qn_t = self.target_model(batch["next_obs"])
+                    qn_o = self.model(batch["next_obs"]) \
+                        if args.double_dqn else None
+                loss, td_abs = fused_td_loss(
+                    q, qn_o, qn_t, batch["action"], batch["reward"],
+                    batch["discount"], **per_kw)
             loss.backward()
             all_reduce_flat(self.flat.flat_grad, average=True)
             if args.max_grad_norm > 0:

@@ -121,3 +121,28 @@ def test_ddppo_single_rank_iteration():
     assert s["steps"] == 16
     s2 = t.train_iteration()
     assert np.isfinite(s2["loss"])
+
+
+def test_apex_c51_image_head_end_to_end():
+    """Categorical (C51) DQN on the Nature-CNN image model (reference
+    declares v_min/v_max/atoms at Atari scale, rl_args.py:221-260)."""
+    from scalerl_amd.runtime.apex import ApexTrainer
+    args = ApexArguments(num_actors=2, envs_per_actor=4, buffer_size=4096,
+                         batch_size=32, warmup_learn_steps=256,
+                         learner_update_times=1, device="cpu", seed=5,
+                         categorical_dqn=True, num_atoms=21,
+                         v_min=-5.0, v_max=5.0,
+                         target_update_frequency=10, publish_interval=5)
+    t = ApexTrainer(args)
+    try:
+        t.start_actors()
+        t.setup_learner()
+        got_loss = False
+        for _ in range(6):
+            s = t.train_iteration()
+            if "loss" in s:
+                got_loss = True
+                assert torch.isfinite(s["loss"])
+        assert got_loss
+    finally:
+        t.shutdown()

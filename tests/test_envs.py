@@ -111,3 +111,66 @@ def test_torch_env_wrapper_dict_protocol():
     out = w.step(torch.tensor(1))
     assert out["episode_step"].item() == 1
     assert out["reward"].item() == 1.0
+
+
+def test_record_episode_statistics_contract():
+    """gymnasium RecordEpisodeStatistics dict contract on terminal steps
+    (reference make_gym_env parity, gym_env.py:6-33)."""
+    from scalerl_amd.envs.registry import make_gym_env
+    env = make_gym_env("CartPole-v1", seed=0)
+    obs, _ = env.reset(seed=0)
+    done_info = None
+    for _ in range(501):
+        obs, r, term, trunc, info = env.step(env.action_space.sample())
+        if term or trunc:
+            done_info = info
+            break
+    assert done_info is not None and "episode" in done_info
+    ep = done_info["episode"]
+    assert ep["l"] >= 1 and ep["r"] == ep["l"]  # CartPole reward = 1/step
+    env.close()
+
+
+def test_record_video_writes_gif(tmp_path):
+    """RecordVideo captures the triggered episode's frames to a GIF."""
+    from scalerl_amd.envs.registry import make_gym_env
+    env = make_gym_env("synthetic-atari", seed=0, capture_video=True,
+                       save_video_dir=str(tmp_path), save_video_name="t")
+    env.reset(seed=0)  # episode 0 triggers (0^3 == 0)
+    for _ in range(40):
+        obs, r, term, trunc, info = env.step(env.action_space.sample())
+        if term or trunc:
+            break
+    env.close()
+    import glob
+    files = glob.glob(str(tmp_path / "t" / "*.gif"))
+    assert files, "no GIF written"
+    from PIL import Image
+    im = Image.open(files[0])
+    assert im.n_frames >= 2
+
+
+def _have_ale() -> bool:
+    try:
+        import gymnasium  # noqa: F401
+        import ale_py  # noqa: F401
+        return True
+    except ImportError:
+        return False
+
+
+@pytest.mark.skipif(not _have_ale(), reason="gymnasium/ALE not installed "
+                    "in this image (no network); runs where they exist")
+def test_real_ale_deepmind_stack_smoke():
+    """Compose the DeepMind wrapper stack over a real ALE env through
+    GymAdapter (reference wrap_deepmind, atari_wrapper.py:277-311)."""
+    from scalerl_amd.envs.registry import make_env
+    env = make_env("ALE/Pong-v5", seed=0, deepmind_wrap=True)
+    obs, _ = env.reset(seed=0)
+    assert obs.shape == (4, 84, 84) and obs.dtype.name == "uint8"
+    for _ in range(20):
+        obs, r, term, trunc, info = env.step(env.action_space.sample())
+        assert obs.shape == (4, 84, 84)
+        if term or trunc:
+            break
+    env.close()
