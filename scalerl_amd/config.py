@@ -130,6 +130,28 @@ class A3CArguments(RLArguments):
 
 
 @dataclass
+class A3CGpuArguments(A3CArguments):
+    """A3C at GPU scale (BASELINE config 2: Pong 42×42, 16 CPU actors +
+    1 MI355X learner) — batched synchronous A2C on the shared
+    actor-learner runtime (reference counterpart: parallel_a3c.py:71-513
+    scaled past hogwild)."""
+
+    algo_name: str = "a3c-gpu"
+    env_id: str = "synthetic-atari"  # ALE Pong when gymnasium is present
+    num_actors: int = _h("CPU actor processes", default=16)
+    envs_per_actor: int = _h("vectorized envs per actor", default=8)
+    rollout_steps: int = 20
+    slots_per_batch: int = _h("rollout slots per learner batch", default=4)
+    gae_lambda: float = 1.0
+    learning_rate: float = 1e-4
+    max_grad_norm: float = 50.0
+    dtype: str = _h("learner compute dtype: bf16 | fp32", default="bf16")
+    total_steps: int = _h("total env steps", default=1_000_000)
+    disable_checkpoint: bool = _h("skip periodic checkpoints", default=False)
+    output_dir: str = _h("checkpoint directory", default="work_dirs/a3c_gpu")
+
+
+@dataclass
 class ImpalaArguments(RLArguments):
     """IMPALA (fields the reference reads in impala_atari.py but never
     declares — SURVEY.md 'Broken-as-shipped' list)."""

@@ -51,3 +51,27 @@ class A3CAtariNet(nn.Module):
         out, state = self.core(x, notdone, state)
         h = out.squeeze(0)
         return self.actor(h), self.critic(h).squeeze(-1), state
+
+    def unroll(self, obs: torch.Tensor, notdone: torch.Tensor, state):
+        """Learner-side unroll: obs [T,B,C,42,42], notdone [T,B] →
+        (logits [T,B,A], values [T,B], state).  Convs batched over T*B,
+        recurrence through the MaskedLSTM core (done-masked)."""
+        T, B = obs.shape[:2]
+        x = obs.flatten(0, 1)
+        x = F.elu(self.conv1(x))
+        x = F.elu(self.conv2(x))
+        x = F.elu(self.conv3(x))
+        x = F.elu(self.conv4(x))
+        x = x.flatten(1).view(T, B, -1)
+        out, state = self.core(x, notdone, state)
+        h = out.flatten(0, 1).float()
+        if h.is_cuda:
+            # heads fp32 under autocast (value/logit noise sensitivity)
+            with torch.autocast(device_type="cuda", enabled=False):
+                logits = self.actor(h)
+                values = self.critic(h).squeeze(-1)
+        else:
+            logits = self.actor(h)
+            values = self.critic(h).squeeze(-1)
+        return (logits.view(T, B, self.num_actions), values.view(T, B),
+                state)

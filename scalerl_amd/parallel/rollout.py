@@ -42,7 +42,8 @@ class RolloutStore:
 
     def __init__(self, num_slots: int, rollout_length: int, envs_per_slot: int,
                  obs_shape: Tuple[int, ...], num_actions: int,
-                 lstm_layers: int = 2, lstm_hidden: int = 0):
+                 lstm_layers: int = 2, lstm_hidden: int = 0,
+                 obs_dtype: torch.dtype = torch.uint8):
         S, T1, E = num_slots, rollout_length + 1, envs_per_slot
         self.num_slots = S
         self.rollout_length = rollout_length
@@ -56,7 +57,7 @@ class RolloutStore:
             t.share_memory_()
             return t
 
-        self.obs = shared((S, T1, E, *obs_shape), torch.uint8)
+        self.obs = shared((S, T1, E, *obs_shape), obs_dtype)
         self.reward = shared((S, T1, E), torch.float32)
         self.done = shared((S, T1, E), torch.bool)
         self.last_action = shared((S, T1, E), torch.int64)
@@ -229,11 +230,18 @@ def build_actor_env(env_spec: dict, actor_id: int):
     env_id = env_spec["env_id"]
     E = env_spec["envs_per_actor"]
     seed = env_spec.get("seed", 0)
+    from ..envs.vec_env import SyncVectorEnv
+    if env_spec.get("a3c_wrap"):
+        # 42×42 normalized A3C preprocessing (a3c_env.create_atari_env)
+        from ..envs.a3c_env import create_atari_env
+        return SyncVectorEnv([
+            (lambda i=i: create_atari_env(
+                env_id, seed=seed * 1000 + actor_id * E + i))
+            for i in range(E)])
     if env_id == "synthetic-atari":
         from ..envs.synthetic import SyntheticAtariVecEnv
         return SyntheticAtariVecEnv(E, seed=seed * 1000 + actor_id)
     from ..envs.registry import make_env
-    from ..envs.vec_env import SyncVectorEnv
     return SyncVectorEnv([
         (lambda i=i: make_env(env_id, seed=seed * 1000 + actor_id * E + i,
                               deepmind_wrap=env_spec.get("deepmind_wrap", True)))

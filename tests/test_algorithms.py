@@ -146,3 +146,24 @@ def test_apex_c51_image_head_end_to_end():
         assert got_loss
     finally:
         t.shutdown()
+
+
+def test_a3c_gpu_trainer_cpu_iteration():
+    """BASELINE config 2 wiring: 42×42 A3CAtariNet + a3c env stack on the
+    batched actor-learner runtime (CPU smoke of the GPU-scale path)."""
+    from scalerl_amd.config import A3CGpuArguments
+    from scalerl_amd.runtime.a3c_gpu import A3CGpuTrainer
+    args = A3CGpuArguments(num_actors=2, envs_per_actor=4, rollout_steps=8,
+                           slots_per_batch=2, device="cpu", seed=7,
+                           disable_checkpoint=True)
+    t = A3CGpuTrainer(args)
+    assert t.obs_shape == (1, 42, 42)
+    try:
+        t.start_actors()
+        t.setup_learner()
+        for _ in range(3):
+            s = t.train_iteration()
+            assert torch.isfinite(s["total_loss"])
+        assert t.global_step == 3 * 8 * (2 * 4)
+    finally:
+        t.shutdown()
