@@ -12,6 +12,18 @@ and weight publication back to the actors.
 
 Single node, one rank per GPU (launched by torch.distributed.run for
 N > 1); weak scaling — per-rank actor count and batch are fixed.
+
+8-rank resource budget (validated CPU-side; driver runs the real SCALE):
+- /dev/shm: each rank's rollout store is num_buffers × slot_bytes, slot ≈
+  (T+1)·E·(4·84·84 u8 + A·4 logits + 28 scalars) ≈ 292 MB at T=80, E=128.
+  Defaults give ~30 slots ≈ 8.8 GB/rank → ~70 GB at 8 ranks.  main()
+  checks statvfs(/dev/shm) and halves envs_per_actor (keeping batch_size
+  divisible) until the projected total fits in 80% of free shm.
+- cores: per-rank actor count is sized from cpu_count()/world minus 2
+  (learner + inference worker): 256 cores / 8 ranks → 24 actors each,
+  ~208 processes total.
+- MIOpen: MIOPEN_FIND_MODE=1 plus a shared MIOPEN_USER_DB_PATH so 8
+  concurrent learners share one find cache instead of 8 find storms.
 """
 
 from __future__ import annotations
@@ -47,8 +59,11 @@ def parse_args():
 
 def main():
     # Consistent MIOpen behavior across boxes: full find (the hybrid default
-    # has been observed to settle on naive bf16 conv kernels on fresh boxes).
+    # has been observed to settle on naive bf16 conv kernels on fresh boxes),
+    # and ONE shared find-cache db across all ranks of a node.
     os.environ.setdefault("MIOPEN_FIND_MODE", "1")
+    os.environ.setdefault("MIOPEN_USER_DB_PATH", "/tmp/scalerl_miopen")
+    os.makedirs(os.environ["MIOPEN_USER_DB_PATH"], exist_ok=True)
     args = parse_args()
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -66,7 +81,8 @@ def main():
 
     if args.num_actors <= 0:
         ncpu = os.cpu_count() or 8
-        avail = max(2, (ncpu - 3 * world) // max(1, world))
+        # per-rank core budget: actors + learner + inference worker
+        avail = max(2, ncpu // max(1, world) - 2)
         if inference == "gpu":
             # actors are pure env-steppers; a handful saturate the
             # inference worker
@@ -74,6 +90,25 @@ def main():
         else:
             # one CPU-inference actor ≈ a few hundred steps/s
             args.num_actors = min(64, avail)
+
+    # /dev/shm budget: shrink envs_per_actor until all ranks' rollout
+    # stores fit in 80% of the free space (see module docstring).
+    try:
+        st = os.statvfs("/dev/shm")
+        shm_free = st.f_bavail * st.f_frsize
+    except OSError:
+        shm_free = None
+    if shm_free is not None:
+        def store_bytes(E):
+            slots = args.num_actors + 2 * max(args.batch_size // E, 1) + 2
+            slot = (args.rollout_length + 1) * E * (4 * 84 * 84 + 6 * 4 + 28)
+            return slots * slot
+        while (args.envs_per_actor > 16
+               and world * store_bytes(args.envs_per_actor) > 0.8 * shm_free):
+            args.envs_per_actor //= 2
+            if rank == 0:
+                print(f"[bench] /dev/shm pressure: envs_per_actor -> "
+                      f"{args.envs_per_actor}", flush=True)
 
     cfg = ImpalaArguments(
         rollout_length=args.rollout_length, batch_size=args.batch_size,

@@ -53,3 +53,25 @@ def test_bench_torchrun_two_ranks_cpu():
     assert j["config"]["parallelism"] == "dp2"
     # whole-job value: 2 ranks × 2 steps × 8×8 env steps / elapsed
     assert j["value"] > 0
+
+
+@pytest.mark.timeout(1200)
+def test_bench_torchrun_eight_ranks_cpu_dryrun():
+    """8-rank process topology dry-run on CPU (tiny shapes): proves the
+    torchrun launch, per-rank actor farms, gloo all-reduce and the timed
+    loop all come up at the driver's SCALE rank count."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+           "--master-port", "29629", BENCH, "--steps", "2", "--warmup", "1",
+           "--rollout-length", "8", "--batch-size", "8", "--envs-per-actor",
+           "4", "--num-actors", "2", "--device", "cpu", "--dtype", "fp32"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=1100,
+                       cwd=REPO, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-3000:])
+    j = _parse_last_json(r.stdout)
+    assert j["config"]["parallelism"] == "dp8"
+    assert j["n_gpus"] == 0  # CPU dry-run
+    assert j["value"] > 0
