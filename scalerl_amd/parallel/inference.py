@@ -103,6 +103,7 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
     from .flat import FlatParams
     from .rollout import pin_tensor
 
+    import os
     torch.manual_seed(seed)
     device = torch.device(f"cuda:{device_index}")
     torch.cuda.set_device(device)
@@ -112,6 +113,10 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
     for t in slots.tensors():
         pin_tensor(t)
     pin_tensor(shared_flat)
+    # bf16 convs/FC for the behavior forward: the recorded behavior logits
+    # are the ones actions are sampled from, so the V-trace correction
+    # stays exact; heads/LSTM run fp32 inside AtariNet regardless.
+    inf_bf16 = bool(os.environ.get("SCALERL_INF_BF16"))
 
     A, E = slots.num_actors, slots.envs_per_actor
     nact = slots.num_actions
@@ -188,7 +193,11 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
                         snap[a] = (h_all[:, a * E:(a + 1) * E].clone(),
                                    c_all[:, a * E:(a + 1) * E].clone())
                 state = (h_all.clone(), c_all.clone())
-            out, new_state = model(inputs, state)
+            if inf_bf16:
+                with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                    out, new_state = model(inputs, state)
+            else:
+                out, new_state = model(inputs, state)
             action = out["action"].view(A, E)
             logits = out["policy_logits"].view(A, E, nact)
             for a in ids:
