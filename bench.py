@@ -119,16 +119,19 @@ def main():
         total_steps=1 << 60, disable_checkpoint=True,
         output_dir="/tmp/scalerl_bench")
 
-    # Build trainer (forks actor processes) BEFORE any CUDA/HIP init.
     trainer = ImpalaTrainer(cfg, device=device)
     trainer.rank = rank
-    trainer.start_actors()
-
-    # Now the device side (CUDA ctx + RCCL process group).
+    if trainer._mp_ctx == "fork":
+        # pure-CPU path: fork actors BEFORE any device init
+        trainer.start_actors()
     if world > 1:
         from scalerl_amd.parallel.dist import init_distributed
         init_distributed("nccl" if use_cuda else "gloo")
+    # GPU path (spawn ctx): device init + hipGraph capture FIRST, then
+    # spawn the inference worker + actors (capture must precede any other
+    # process submitting on this GPU — see ImpalaTrainer.setup_learner)
     trainer.setup_learner()
+    trainer.start_actors()
 
     def barrier_sync():
         if world > 1:

@@ -230,10 +230,13 @@ _USE_SEQ = None
 
 
 def _use_seq_path() -> bool:
+    """C++-driven sequence loop (hardware-validated r2: oracle tests pass,
+    25.5 vs 28.2 ms/iter at B=256 on the learner micro).  Default ON on
+    GPU; SCALERL_LSTM_SEQ=0 falls back to the per-step Python path."""
     global _USE_SEQ
     if _USE_SEQ is None:
         import os
-        _USE_SEQ = bool(os.environ.get("SCALERL_LSTM_SEQ"))
+        _USE_SEQ = os.environ.get("SCALERL_LSTM_SEQ", "1") != "0"
     return _USE_SEQ
 
 

@@ -206,7 +206,47 @@ class ImpalaTrainer:
         self.autocast_dtype = (torch.bfloat16 if args.dtype == "bf16" and
                                self.device.type == "cuda" else None)
 
+        # hipGraph capture happens HERE — before the inference worker
+        # process exists.  Round-1 finding: concurrent HIP submissions from
+        # another process during stream capture abort the other process's
+        # HSA queue on ROCm 7.2, so capture-from-a-synthetic-batch precedes
+        # start_actors() (callers: bench.py, train()).  The pause handshake
+        # remains as a fallback for the lazy capture path.
+        if self.use_graph and self.device.type == "cuda" and not self._started:
+            self._capture_graph()
+        self._setup_remote_server()
+
+    def _capture_graph(self) -> None:
+        from .graphed import GraphedImpalaStep
+        args = self.args
+        T, B = args.rollout_length, args.batch_size
+        dev = self.device
+        g = torch.Generator(device="cpu").manual_seed(0)
+        synth = {
+            "obs": torch.randint(0, 256, (T + 1, B, *self.obs_shape),
+                                 dtype=torch.uint8, device=dev),
+            "reward": torch.randn(T + 1, B, generator=g).to(dev),
+            "done": (torch.rand(T + 1, B, generator=g) < 0.01).to(dev),
+            "last_action": torch.randint(
+                0, self.num_actions, (T + 1, B), generator=g).to(dev),
+            "action": torch.randint(
+                0, self.num_actions, (T + 1, B), generator=g).to(dev),
+            "logits": torch.randn(T + 1, B, self.num_actions,
+                                  generator=g).to(dev),
+        }
+        if args.use_lstm:
+            synth["core_state"] = torch.zeros(
+                2, 2, B, self.store.lstm_hidden, device=dev)
+        self.log.info("capturing learner step into a hipGraph "
+                      "(pre-actor-start) …")
+        self._graphed = GraphedImpalaStep(
+            self.learner_model, self.flat.flat_grad, self._loss_kwargs(),
+            synth, args.use_lstm, self.autocast_dtype,
+            args.reward_clipping == "abs_one", args.discounting)
+
+    def _setup_remote_server(self) -> None:
         # optional TCP server for remote-node actor farms
+        args = self.args
         self.remote_server = None
         if args.remote_actor_slots > 0:
             import queue as _q
@@ -397,8 +437,14 @@ class ImpalaTrainer:
     # -- driver ------------------------------------------------------------
     def train(self) -> None:
         args = self.args
-        self.start_actors()
-        self.setup_learner()
+        if self._mp_ctx == "spawn":
+            # device init first so graph capture precedes the inference
+            # worker; spawned children don't inherit the HIP runtime
+            self.setup_learner()
+            self.start_actors()
+        else:
+            self.start_actors()
+            self.setup_learner()
         ckpt_path = args.checkpoint_path or os.path.join(
             args.output_dir, "model.tar")
         last_ckpt = time.time()

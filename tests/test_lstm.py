@@ -88,8 +88,6 @@ def test_masked_lstm_gpu_matches_cpu():
 
 
 @pytest.mark.gpu
-@pytest.mark.skipif(not __import__("os").environ.get("SCALERL_EXPERIMENTAL"),
-                    reason="C++-loop LSTM path pending hardware validation")
 def test_masked_lstm_seq_path_matches_python_path():
     """SCALERL_LSTM_SEQ C++-driven loop vs the per-step Python path."""
     import scalerl_amd.ops.lstm as lstm_mod
