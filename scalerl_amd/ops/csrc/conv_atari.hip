@@ -62,9 +62,10 @@ extern "C" int mfma_selftest(const float* A, const float* B, float* D,
   return 0;
 }
 
-// ---- generic implicit-GEMM forward conv (compile-time shape) ----
-// Block: 4 waves in a 2(M)x2(N) arrangement -> 32x32 output tile.
-// IN_U8: input is uint8, normalized by /255 on load (conv1).
+// Forward kernels live in conv_fwd.hip (v2, LDS-staged).  The v1 generic
+// gather-per-lane template below is retained ONLY as the shape reference
+// for the backward kernels' masks; it is compiled out.
+#if 0
 template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
           int KOUT, bool IN_U8, typename in_t>
 __global__ __launch_bounds__(256) void conv_fwd_kernel(
@@ -157,12 +158,11 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
     return 0;                                                                 \
   }
 
-DEF_CONV(atari_conv1_fwd_u8, 4, 8, 8, 4, 84, 84, 20, 20, 32, true,
-         unsigned char)
-DEF_CONV(atari_conv1_fwd_bf16, 4, 8, 8, 4, 84, 84, 20, 20, 32, false, bf16_t)
-DEF_CONV(atari_conv2_fwd, 32, 4, 4, 2, 20, 20, 9, 9, 64, false, bf16_t)
-DEF_CONV(atari_conv3_fwd, 64, 3, 3, 1, 9, 9, 7, 7, 64, false, bf16_t)
+#endif  // v1 forward (superseded by conv_fwd.hip)
 
+// Backward kernels live in conv_bwd.hip (v2).  v1 retained for
+// reference only, compiled out.
+#if 0
 // ---- backward: weight gradient ---------------------------------------
 // dw[k_out, kdim] = sum_pixels dy[pixel, k_out] * im2col(x)[pixel, kdim]
 // GEMM: rows = K_OUT, cols = KDIM, reduce over M = N*OH*OW pixels.
@@ -328,3 +328,5 @@ DEF_WGRAD(atari_conv2_wgrad, 32, 4, 4, 2, 20, 20, 9, 9, 64, false, bf16_t)
 DEF_WGRAD(atari_conv3_wgrad, 64, 3, 3, 1, 9, 9, 7, 7, 64, false, bf16_t)
 DEF_DGRAD(atari_conv2_dgrad, 32, 4, 4, 2, 20, 20, 9, 9, 64)
 DEF_DGRAD(atari_conv3_dgrad, 64, 3, 3, 1, 9, 9, 7, 7, 64)
+
+#endif  // v1 backward (superseded by conv_bwd.hip)

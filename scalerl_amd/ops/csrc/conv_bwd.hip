@@ -1,0 +1,263 @@
+// Backward (wgrad + dgrad) MFMA kernels for the Atari encoder convs — v2,
+// same LDS-staging design as conv_fwd.hip (the v1 gather-per-lane versions
+// measured far behind MIOpen and conv3 dgrad failed its oracle on HW).
+//
+// wgrad: dw[ko][k] = sum_px dy[px,ko] * im2col(x)[px,k]
+//   - each block owns the FULL dw tile-set (conv1) or a KDIM quarter
+//     (conv2/3), loops over a strided subset of images with x and dy staged
+//     in LDS, accumulates in registers, atomicAdds once at the end;
+//   - A (dy) fragments are 16B-contiguous pixel runs (1 ds_read_b128);
+//     B (im2col) fragments are scalar LDS gathers.
+// dgrad: dx[ipx][c] = sum_{ko,ky,kx valid} dy[n,ko,oy,ox] * w[ko,c,ky,kx]
+//   - per-image blocks stage dy in LDS (masked scalar gathers), weights
+//     come from global (L1-resident), dx staged through LDS for one
+//     coalesced vectorized writeback.
+//
+// Fragment maps as in conv_fwd.hip (validated by mfma_selftest).
+
+#include "common.h"
+
+typedef __bf16 bf16_t;
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef bf16_t bf16x8 __attribute__((ext_vector_type(8)));
+
+// ---------------------------------------------------------------- wgrad --
+
+// conv2/conv3 wgrad share a template: block owns a KDIM quarter.
+//  CONVID 2: x [N,32,20,20], dy [N,64,9,9],  dw [64,512], quarter = 8 ch
+//  CONVID 3: x [N,64,9,9],   dy [N,64,7,7],  dw [64,576], quarter = 16 ch
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT, int CQ, int NTQ, int TPW, bool IN_U8 = false,
+          typename in_t = bf16_t>
+__global__ __launch_bounds__(256) void convN_wgrad_v2(
+    const in_t* __restrict__ input, const bf16_t* __restrict__ dout,
+    float* __restrict__ dweight, int batch) {
+  constexpr int KDIM = C * KH * KW;
+  constexpr int KQ = CQ * KH * KW;          // k-columns per quarter
+  constexpr int IPITCH = ((IW + 3) & ~3);
+  constexpr int MPX = OH * OW;
+  constexpr int PXP = ((MPX + 31) & ~31);   // pitch AND k-loop bound
+  constexpr int MT = KOUT / 16;
+
+  __shared__ bf16_t img[CQ * IH * IPITCH];
+  __shared__ bf16_t dy[KOUT * PXP];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g = lane >> 4, lr = lane & 15;
+  const int q = blockIdx.y;                 // KDIM quarter
+  const int c0 = q * CQ;
+
+  f32x4 acc[TPW];
+  #pragma unroll
+  for (int i = 0; i < TPW; ++i) acc[i] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  for (int n = blockIdx.x; n < batch; n += gridDim.x) {
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < CQ * IH * IW; idx += blockDim.x) {
+      const int c = idx / (IH * IW);
+      const int rem = idx % (IH * IW);
+      const in_t raw = input[(((long)n * C + c0 + c) * IH * IW) + rem];
+      img[(c * IH + rem / IW) * IPITCH + rem % IW] =
+          (bf16_t)(IN_U8 ? (float)raw * (1.0f / 255.0f) : (float)raw);
+    }
+    for (int idx = threadIdx.x; idx < KOUT * PXP; idx += blockDim.x) {
+      const int ko = idx / PXP, px = idx % PXP;
+      dy[idx] = (px < MPX)
+          ? dout[((long)n * KOUT + ko) * MPX + px] : (bf16_t)0.f;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ti = 0; ti < TPW; ++ti) {
+      const int t = wave + ti * 4;
+      const int mt = t / NTQ, nt = t % NTQ;
+      const int ko = mt * 16 + lr;
+      const int kq = nt * 16 + lr;          // column within quarter
+      const int c = kq / (KH * KW);
+      const int r = kq % (KH * KW);
+      const int ky = r / KW, kx = r % KW;
+      #pragma unroll 2
+      for (int pt = 0; pt < PXP / 32; ++pt) {
+        const int p0 = pt * 32 + g * 8;
+        const bf16x8 a = *(const bf16x8*)&dy[ko * PXP + p0];
+        bf16x8 b;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int p = p0 + j;
+          const int oy = p / OW, ox = p % OW;
+          bf16_t v = (bf16_t)0.f;
+          if (p < MPX)
+            v = img[(c * IH + oy * STRIDE + ky) * IPITCH + ox * STRIDE + kx];
+          b[j] = v;
+        }
+        acc[ti] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ti],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  #pragma unroll
+  for (int ti = 0; ti < TPW; ++ti) {
+    const int t = wave + ti * 4;
+    const int mt = t / NTQ, nt = t % NTQ;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int col = q * KQ + nt * 16 + lr;
+      const int row = mt * 16 + g * 4 + r;
+      atomicAdd(&dweight[(long)row * KDIM + col], acc[ti][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- dgrad --
+// Template over shape; IMGS images per block.  M = IMGS*IH*IW input px,
+// N = C, K = KOUT*KH*KW with validity-masked dy gathers from LDS.
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT, int IMGS>
+__global__ __launch_bounds__(256) void convN_dgrad_v2(
+    const bf16_t* __restrict__ dout,    // [N, KOUT, OH, OW]
+    const bf16_t* __restrict__ weight,  // [KOUT, C, KH, KW]
+    bf16_t* __restrict__ dinput,        // [N, C, IH, IW]
+    int batch) {
+  constexpr int RDIM = KOUT * KH * KW;
+  constexpr int OPX = OH * OW;
+  constexpr int OPP = ((OPX + 7) & ~7);
+  constexpr int IPX = IH * IW;
+  constexpr int MPX = IMGS * IPX;
+  constexpr int MT = (MPX + 15) / 16, NT = C / 16;
+
+  __shared__ bf16_t dy[IMGS * KOUT * OPP];
+  __shared__ bf16_t dx[C * MPX];
+
+  const int n0 = blockIdx.x * IMGS;
+  const int n_here = min(IMGS, batch - n0);
+  for (int idx = threadIdx.x; idx < n_here * KOUT * OPX;
+       idx += blockDim.x) {
+    const int i = idx / (KOUT * OPX);
+    const int rem = idx % (KOUT * OPX);
+    const int ko = rem / OPX, px = rem % OPX;
+    dy[(i * KOUT + ko) * OPP + px] =
+        dout[((long)(n0 + i) * KOUT + ko) * OPX + px];
+  }
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g = lane >> 4, lr = lane & 15;
+
+  for (int t = wave; t < MT * NT; t += 4) {
+    const int mt = t / NT, nt = t % NT;
+    const int row = mt * 16 + lr;
+    const int i = row / IPX, px = row % IPX;
+    const bool ok = row < IPX * n_here;
+    const int iy = px / IW, ix = px % IW;
+    const int c = nt * 16 + lr;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int kt = 0; kt < (RDIM + 31) / 32; ++kt) {
+      const int k0 = kt * 32 + g * 8;
+      bf16x8 a, b;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int k = k0 + j;
+        bf16_t av = (bf16_t)0.f, bv = (bf16_t)0.f;
+        if (k < RDIM) {
+          const int ko = k / (KH * KW);
+          const int r = k % (KH * KW);
+          const int ky = r / KW, kx = r % KW;
+          if (ok) {
+            const int ty = iy - ky, tx = ix - kx;
+            if (ty >= 0 && tx >= 0 && ty % STRIDE == 0 &&
+                tx % STRIDE == 0) {
+              const int oy = ty / STRIDE, ox = tx / STRIDE;
+              if (oy < OH && ox < OW)
+                av = dy[(i * KOUT + ko) * OPP + oy * OW + ox];
+            }
+          }
+          bv = weight[(((long)ko * C + c) * KH + ky) * KW + kx];
+        }
+        a[j] = av;
+        b[j] = bv;
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int rw = mt * 16 + g * 4 + r;
+      if (rw < MPX)
+        dx[(nt * 16 + lr) * MPX + rw] = (bf16_t)acc[r];
+    }
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < n_here * C * IPX; idx += blockDim.x) {
+    const int i = idx / (C * IPX);
+    const int rem = idx % (C * IPX);
+    const int c = rem / IPX, px = rem % IPX;
+    dinput[((long)(n0 + i) * C + c) * IPX + px] =
+        dx[c * MPX + i * IPX + px];
+  }
+}
+
+// ---- exported entry points (same ABI as v1; `split` = grid-size hint) ---
+static inline int wgrad_grid(long units) {
+  long g = units < 768 ? units : 768;
+  return (int)(g < 1 ? 1 : g);
+}
+
+extern "C" int atari_conv1_wgrad_u8(const void* in, const void* dout,
+                                    float* dw, long batch, long split,
+                                    hipStream_t stream) {
+  (void)split;
+  // 2 mt x 4 nt = 8 tiles -> 2 per wave; quarter = 1 input channel
+  hipLaunchKernelGGL(
+      (convN_wgrad_v2<4, 8, 8, 4, 84, 84, 20, 20, 32, 1, 4, 2, true,
+                      unsigned char>),
+      dim3(wgrad_grid(batch), 4), dim3(256), 0, stream,
+      (const unsigned char*)in, (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv2_wgrad(const void* in, const void* dout, float* dw,
+                                 long batch, long split, hipStream_t stream) {
+  (void)split;
+  // 4 mt x 4 nt = 16 tiles -> 4 per wave; quarter = 4 input channels
+  hipLaunchKernelGGL(
+      (convN_wgrad_v2<32, 4, 4, 2, 20, 20, 9, 9, 64, 4, 4, 4>),
+      dim3(wgrad_grid(batch), 8), dim3(256), 0, stream, (const bf16_t*)in,
+      (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv3_wgrad(const void* in, const void* dout, float* dw,
+                                 long batch, long split, hipStream_t stream) {
+  (void)split;
+  // 4 mt x 9 nt = 36 tiles -> 9 per wave; quarter = 16 input channels
+  hipLaunchKernelGGL(
+      (convN_wgrad_v2<64, 3, 3, 1, 9, 9, 7, 7, 64, 16, 9, 9>),
+      dim3(wgrad_grid(batch), 4), dim3(256), 0, stream, (const bf16_t*)in,
+      (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv2_dgrad(const void* dout, const void* w, void* din,
+                                 long batch, hipStream_t stream) {
+  hipLaunchKernelGGL((convN_dgrad_v2<32, 4, 4, 2, 20, 20, 9, 9, 64, 1>),
+                     dim3((unsigned)batch), dim3(256), 0, stream,
+                     (const bf16_t*)dout, (const bf16_t*)w, (bf16_t*)din,
+                     (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv3_dgrad(const void* dout, const void* w, void* din,
+                                 long batch, hipStream_t stream) {
+  hipLaunchKernelGGL((convN_dgrad_v2<64, 3, 3, 1, 9, 9, 7, 7, 64, 2>),
+                     dim3((unsigned)((batch + 1) / 2)), dim3(256), 0, stream,
+                     (const bf16_t*)dout, (const bf16_t*)w, (bf16_t*)din,
+                     (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
