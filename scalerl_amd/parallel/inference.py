@@ -192,7 +192,10 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
                     if slots.want_state[a]:
                         snap[a] = (h_all[:, a * E:(a + 1) * E].clone(),
                                    c_all[:, a * E:(a + 1) * E].clone())
-                state = (h_all.clone(), c_all.clone())
+                # no clone: the LSTM builds fresh state tensors, and only
+                # requesting actors' slices are copied back below — cloning
+                # h/c here cost ~50 MB of D2D per round at A=24,E=128
+                state = (h_all, c_all)
             if inf_bf16:
                 with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
                     out, new_state = model(inputs, state)

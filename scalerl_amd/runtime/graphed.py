@@ -67,17 +67,29 @@ class GraphedImpalaStep:
             total.backward()
             return total.detach(), comps
 
-        # torch-required warmup on a side stream before capture
-        s = torch.cuda.Stream(device=dev)
-        s.wait_stream(torch.cuda.current_stream(dev))
-        with torch.cuda.stream(s):
-            for _ in range(warmup_iters):
-                run_step()
-        torch.cuda.current_stream(dev).wait_stream(s)
+        # The C++ LSTM sequence loop (ops/csrc/lstm_seq.hip) drives raw
+        # rocBLAS calls that are not graph-capture-safe (r2 call-2 finding:
+        # HSA exception during capture with no other process on the GPU).
+        # Inside a captured graph the per-step path's Python launch
+        # overhead is paid once at capture time only, so force it here.
+        from ..ops import lstm as _lstm_mod
+        prev_seq = _lstm_mod._USE_SEQ
+        _lstm_mod._USE_SEQ = False
+        try:
+            # torch-required warmup on a side stream before capture
+            s = torch.cuda.Stream(device=dev)
+            s.wait_stream(torch.cuda.current_stream(dev))
+            with torch.cuda.stream(s):
+                for _ in range(warmup_iters):
+                    run_step()
+            torch.cuda.current_stream(dev).wait_stream(s)
 
-        self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
-            self.total, self.comps = run_step()
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph,
+                                  capture_error_mode="thread_local"):
+                self.total, self.comps = run_step()
+        finally:
+            _lstm_mod._USE_SEQ = prev_seq
 
     def run(self, batch: Dict[str, torch.Tensor]) -> Tuple[torch.Tensor, torch.Tensor]:
         for k, dst in self.static.items():
