@@ -115,3 +115,118 @@ def test_dgrad_formula_matches_torch(layer):
     (out * dy).sum().backward()
     got = emu_dgrad(dy.numpy(), w.numpy(), stride, in_shape)
     np.testing.assert_allclose(got, x.grad.numpy(), rtol=1e-6, atol=1e-8)
+
+
+# ---- v2 kernels (csrc/conv_fwd.hip): exact per-lane/per-fragment emulation
+# of the LDS-staged forward index math at the REAL shapes, vs F.conv2d.
+# (The v1 emulations above cover the backward kernels' gather formulas,
+# which conv_bwd.hip reuses.)
+
+def _emulate_conv1_v2(x_u8, w, bias):
+    N = x_u8.shape[0]
+    C, IH, IW, OW, OYB, IYB, PITCH, MPX, KDIM = 4, 84, 84, 20, 10, 44, 96, 200, 256
+    out = np.zeros((N, 32, 20, 20), dtype=np.float32)
+    xb = torch.from_numpy(x_u8.astype(np.float32) / 255.0) \
+        .to(torch.bfloat16).float().numpy()
+    wb = torch.from_numpy(w.reshape(32, KDIM)).to(torch.bfloat16).float().numpy()
+    for n in range(N):
+        for ob in range(2):
+            oy0 = ob * OYB
+            img = np.zeros((C, IYB, PITCH), dtype=np.float32)
+            img[:, :, :IW] = xb[n][:, oy0 * 4:oy0 * 4 + IYB, :]
+            imgf = img.reshape(-1)
+            out_lds = np.zeros((32, MPX), dtype=np.float32)
+            for t in range(((MPX + 15) // 16) * 2):
+                mt, nt = t >> 1, t & 1
+                acc = np.zeros((16, 16), dtype=np.float32)
+                for kt in range(KDIM // 32):
+                    A = np.zeros((16, 32), dtype=np.float32)
+                    B = np.zeros((32, 16), dtype=np.float32)
+                    for lr in range(16):
+                        px = mt * 16 + lr
+                        if px < MPX:
+                            oy_l, ox = px // OW, px % OW
+                            for g in range(4):
+                                k0 = kt * 32 + g * 8
+                                c, ky = k0 >> 6, (k0 & 63) >> 3
+                                off = (c * IYB + oy_l * 4 + ky) * PITCH + ox * 4
+                                A[lr, g * 8:(g + 1) * 8] = imgf[off:off + 8]
+                        ch = nt * 16 + lr
+                        for g in range(4):
+                            k0 = kt * 32 + g * 8
+                            B[g * 8:(g + 1) * 8, lr] = wb[ch, k0:k0 + 8]
+                    acc += A @ B
+                for lr in range(16):
+                    for g in range(4):
+                        for r in range(4):
+                            row = mt * 16 + g * 4 + r
+                            if row < MPX:
+                                out_lds[nt * 16 + lr, row] = acc[g * 4 + r, lr]
+            for ch in range(32):
+                v = np.maximum(out_lds[ch] + bias[ch], 0.0)
+                out[n, ch].reshape(-1)[oy0 * OW:oy0 * OW + MPX] = v
+    return out
+
+
+def test_conv1_v2_fwd_formula_matches_torch():
+    np.random.seed(0)
+    x = np.random.randint(0, 256, (2, 4, 84, 84)).astype(np.uint8)
+    w = (np.random.randn(32, 4, 8, 8) * 0.1).astype(np.float32)
+    b = np.random.randn(32).astype(np.float32)
+    xt = torch.from_numpy(x.astype(np.float32) / 255.0).to(torch.bfloat16).float()
+    wt = torch.from_numpy(w).to(torch.bfloat16).float()
+    ref = F.relu(F.conv2d(xt, wt, torch.from_numpy(b), stride=4)).numpy()
+    got = _emulate_conv1_v2(x, w, b)
+    np.testing.assert_allclose(got, ref, rtol=2e-2, atol=2e-2)
+
+
+def test_conv2_v2_fwd_formula_matches_torch():
+    np.random.seed(1)
+    N, C, IH, IW, OW, OH = 2, 32, 20, 20, 9, 9
+    PITCH, MPX, KDIM, KOUT = 24, 81, 512, 64
+    x = np.random.randn(N, C, IH, IW).astype(np.float32)
+    w = (np.random.randn(KOUT, C, 4, 4) * 0.1).astype(np.float32)
+    b = np.random.randn(KOUT).astype(np.float32)
+    xb = torch.from_numpy(x).to(torch.bfloat16).float().numpy()
+    wb = torch.from_numpy(w.reshape(KOUT, KDIM)).to(torch.bfloat16).float().numpy()
+    ref = F.relu(F.conv2d(torch.from_numpy(xb),
+                          torch.from_numpy(wb.reshape(KOUT, C, 4, 4)),
+                          torch.from_numpy(b), stride=2)).numpy()
+    out = np.zeros((N, KOUT, OH, OW), dtype=np.float32)
+    for n in range(N):
+        img = np.zeros((C, IH, PITCH), dtype=np.float32)
+        img[:, :, :IW] = xb[n]
+        imgf = img.reshape(-1)
+        out_lds = np.zeros((KOUT, MPX), dtype=np.float32)
+        MT, NT = (MPX + 15) // 16, KOUT // 16
+        for t in range(MT * NT):
+            mt, nt = t // NT, t % NT
+            acc = np.zeros((16, 16), dtype=np.float32)
+            for kt in range(KDIM // 32):
+                A = np.zeros((16, 32), dtype=np.float32)
+                B = np.zeros((32, 16), dtype=np.float32)
+                for lr in range(16):
+                    px = mt * 16 + lr
+                    if px < MPX:
+                        oy, ox = px // OW, px % OW
+                        for g in range(4):
+                            k0 = kt * 32 + g * 8
+                            c, ky = k0 >> 4, (k0 & 15) >> 2
+                            base = (c * IH + oy * 2 + ky) * PITCH + ox * 2
+                            A[lr, g * 8:g * 8 + 4] = imgf[base:base + 4]
+                            A[lr, g * 8 + 4:g * 8 + 8] = \
+                                imgf[base + PITCH:base + PITCH + 4]
+                    ch = nt * 16 + lr
+                    for g in range(4):
+                        k0 = kt * 32 + g * 8
+                        B[g * 8:(g + 1) * 8, lr] = wb[ch, k0:k0 + 8]
+                acc += A @ B
+            for lr in range(16):
+                for g in range(4):
+                    for r in range(4):
+                        row = mt * 16 + g * 4 + r
+                        if row < MPX:
+                            out_lds[nt * 16 + lr, row] = acc[g * 4 + r, lr]
+        for ch in range(KOUT):
+            out[n, ch].reshape(-1)[:] = np.maximum(out_lds[ch] + b[ch], 0.0)
+    np.testing.assert_allclose(out, ref, rtol=2e-2, atol=2e-2)
