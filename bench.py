@@ -43,7 +43,9 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--rollout-length", type=int, default=80)
     p.add_argument("--use-graph", type=int, default=0,
-                   help="hipGraph-capture the learner step (experimental)")
+                   help="hipGraph-capture the learner step")
+    p.add_argument("--double-buffer", type=int, default=0,
+                   help="two env groups per actor (SEED latency hiding)")
     p.add_argument("--batch-size", type=int, default=256)
     p.add_argument("--envs-per-actor", type=int, default=128)
     p.add_argument("--num-actors", type=int, default=0,
@@ -115,6 +117,7 @@ def main():
         envs_per_actor=args.envs_per_actor, num_actors=args.num_actors,
         use_lstm=bool(args.use_lstm), device=device, dtype=args.dtype,
         use_graph=bool(args.use_graph),
+        actor_double_buffer=bool(args.double_buffer),
         inference=inference, seed=1234 + rank,
         total_steps=1 << 60, disable_checkpoint=True,
         output_dir="/tmp/scalerl_bench")

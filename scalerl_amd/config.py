@@ -185,9 +185,12 @@ class ImpalaArguments(RLArguments):
     remote_publish_interval: int = _h("learn steps between TCP weight "
                                       "publications", default=10)
     dtype: str = _h("learner compute dtype: bf16 | fp32", default="bf16")
-    use_graph: bool = _h("hipGraph-capture the learner step "
-                         "(experimental: conflicts with the GPU inference "
-                         "worker on ROCm 7.2)", default=False)
+    use_graph: bool = _h("hipGraph-capture the learner step (capture "
+                         "happens before the inference worker starts)",
+                         default=False)
+    actor_double_buffer: bool = _h(
+        "two env groups per actor, interleaved to hide the GPU-inference "
+        "round latency (SEED-style; gpu inference only)", default=False)
 
 
 @dataclass

@@ -70,7 +70,11 @@ class InferenceSlots:
 
 
 class RemotePolicy:
-    """Actor-side client: request an inference round from the worker."""
+    """Actor-side client: request an inference round from the worker.
+
+    Split request/wait API so a double-buffered actor can overlap one env
+    group's inference round with the other group's env stepping (SEED-RL's
+    latency-hiding pattern); ``__call__`` is the blocking convenience."""
 
     def __init__(self, actor_id: int, slots: InferenceSlots, req_q, sem):
         self.aid = actor_id
@@ -78,8 +82,8 @@ class RemotePolicy:
         self.req_q = req_q
         self.sem = sem
 
-    def __call__(self, obs, reward, done, last_action,
-                 want_state: bool = False):
+    def request(self, obs, reward, done, last_action,
+                want_state: bool = False) -> None:
         a = self.aid
         s = self.slots
         s.obs[a].copy_(obs)
@@ -87,11 +91,21 @@ class RemotePolicy:
         s.done[a].copy_(done)
         s.last_action[a].copy_(last_action)
         s.want_state[a] = 1 if want_state else 0
+        self._want_state = want_state
         self.req_q.put(a)
+
+    def wait(self):
+        a = self.aid
+        s = self.slots
         self.sem.acquire()
-        state = s.core_state[a].clone() if (want_state and
+        state = s.core_state[a].clone() if (self._want_state and
                                             s.core_state is not None) else None
         return s.action[a].clone(), s.logits[a].clone(), state
+
+    def __call__(self, obs, reward, done, last_action,
+                 want_state: bool = False):
+        self.request(obs, reward, done, last_action, want_state)
+        return self.wait()
 
 
 def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlots,

@@ -224,6 +224,61 @@ def run_rollout(state: ActorState, store: RolloutStore, slot: int) -> int:
     return T * E
 
 
+def write_row0(state: ActorState, store: RolloutStore, slot: int) -> None:
+    store.obs[slot, 0].copy_(state.obs)
+    store.reward[slot, 0].copy_(state.reward)
+    store.done[slot, 0].copy_(state.done)
+    store.last_action[slot, 0].copy_(state.last_action)
+    store.episode_return[slot, 0].copy_(state.episode_return)
+
+
+@torch.no_grad()
+def run_rollout_db(states, store: RolloutStore, slots) -> int:
+    """Double-buffered rollout: two env groups per actor, each with its own
+    slot, interleaved so one group's env stepping + slot writes overlap the
+    other group's in-flight inference round (SEED-RL latency hiding; the
+    single-group path pays the full round latency every step).
+    Policies must expose request()/wait() (RemotePolicy)."""
+    T = store.rollout_length
+    E = store.envs_per_slot
+    want0 = store.core_state is not None
+    for g in (0, 1):
+        write_row0(states[g], store, slots[g])
+        st = states[g]
+        st.policy.request(st.obs, st.reward, st.done, st.last_action,
+                          want_state=want0)
+    for t in range(T):
+        for g in (0, 1):
+            st = states[g]
+            slot = slots[g]
+            action, logits, snap = st.policy.wait()
+            if snap is not None:
+                store.core_state[slot].copy_(snap)
+            store.action[slot, t].copy_(action)
+            store.logits[slot, t].copy_(logits)
+
+            obs_np, rew_np, done_np = st.env.step(action.numpy())
+            st.episode_return += torch.from_numpy(rew_np)
+            ep_ret = st.episode_return.clone()
+            done_t = torch.from_numpy(done_np)
+            st.episode_return[done_t] = 0.0
+            st.obs = torch.from_numpy(np.ascontiguousarray(obs_np))
+            st.reward = torch.from_numpy(rew_np)
+            st.done = done_t
+            st.last_action = action
+
+            row = t + 1
+            store.obs[slot, row].copy_(st.obs)
+            store.reward[slot, row].copy_(st.reward)
+            store.done[slot, row].copy_(st.done)
+            store.last_action[slot, row].copy_(action)
+            store.episode_return[slot, row].copy_(ep_ret)
+            if t + 1 < T:
+                st.policy.request(st.obs, st.reward, st.done,
+                                  st.last_action)
+    return 2 * T * E
+
+
 def build_actor_env(env_spec: dict, actor_id: int):
     """Construct an actor's vectorized env from a picklable spec
     (module-level so spawned processes can build it)."""
@@ -289,6 +344,55 @@ def actor_loop(actor_id: int, env_spec: dict, store: RolloutStore,
                     try:
                         episode_queue.put_nowait(rets.tolist())
                     except queue.Full:
+                        pass
+    except KeyboardInterrupt:
+        pass
+
+
+def actor_loop_db(actor_id: int, env_spec: dict, store: RolloutStore,
+                  free_q, full_q, stop_event, step_counter,
+                  inf_slots=None, inf_req_q=None, inf_sems=(),
+                  row_ids=(), episode_queue=None, seed: int = 0,
+                  torch_threads: int = 1):
+    """Double-buffered actor process main: two env groups (rows
+    ``row_ids`` in the inference slots), each filling its own rollout slot,
+    interleaved via :func:`run_rollout_db`.  Poison pill: None on free_q."""
+    import queue as _q
+    from .inference import RemotePolicy
+    torch.manual_seed(seed + actor_id)
+    torch.set_num_threads(torch_threads)
+    states = []
+    for g in (0, 1):
+        spec = dict(env_spec)
+        spec["seed"] = env_spec.get("seed", 0) + 500_000 * g
+        env = build_actor_env(spec, actor_id)
+        policy = RemotePolicy(row_ids[g], inf_slots, inf_req_q, inf_sems[g])
+        states.append(ActorState(env, policy, store.envs_per_slot))
+    try:
+        while not stop_event.is_set():
+            slot_a = free_q.get()
+            if slot_a is None:
+                break
+            slot_b = free_q.get()
+            if slot_b is None:
+                break
+            steps = run_rollout_db(states, store, (slot_a, slot_b))
+            full_q.put(slot_a)
+            full_q.put(slot_b)
+            with step_counter.get_lock():
+                step_counter.value += steps
+            if episode_queue is not None:
+                rets = []
+                for slot in (slot_a, slot_b):
+                    mask = store.done[slot, 1:].numpy()
+                    if mask.any():
+                        rets.extend(
+                            store.episode_return[slot, 1:].numpy()[mask]
+                            .tolist())
+                if rets:
+                    try:
+                        episode_queue.put_nowait(rets)
+                    except _q.Full:
                         pass
     except KeyboardInterrupt:
         pass

@@ -70,11 +70,20 @@ class ImpalaTrainer:
             f"batch_size ({args.batch_size}) must be a multiple of " \
             f"envs_per_actor ({E})"
         self.slots_per_batch = args.batch_size // E
-        # enough slots that every actor can have one in flight plus two
-        # batches queued; 2x actors would double host shm for no throughput
-        # (8 ranks x 24 actors x 290 MB slots must fit /dev/shm)
+        # double-buffered actors run two env groups, so each holds TWO
+        # slots in flight (gpu-inference only; see actor_loop_db)
+        effective_inference = args.inference
+        if effective_inference == "gpu" and self.device.type != "cuda":
+            effective_inference = "cpu"
+        self.double_buffer = bool(args.actor_double_buffer) and \
+            effective_inference == "gpu"
+        per_actor_slots = 2 if self.double_buffer else 1
+        # enough slots that every actor can have its group(s) in flight
+        # plus two batches queued; more would grow host shm for no
+        # throughput (8 ranks x 24 actors x 290 MB slots must fit /dev/shm)
         num_buffers = args.num_buffers or (
-            args.num_actors + 2 * self.slots_per_batch + 2)
+            per_actor_slots * args.num_actors +
+            2 * self.slots_per_batch + 2)
         self.local_slots = num_buffers
         num_buffers += args.remote_actor_slots  # reserved tail ids
 
@@ -116,11 +125,12 @@ class ImpalaTrainer:
             self.inference = "cpu"
         self.inference_proc: Optional[mp.Process] = None
         if self.inference == "gpu":
+            rows = args.num_actors * (2 if self.double_buffer else 1)
             self.inf_slots = InferenceSlots(
-                args.num_actors, E, self.obs_shape, self.num_actions,
+                rows, E, self.obs_shape, self.num_actions,
                 lstm_layers=2, lstm_hidden=lstm_hidden)
             self.inf_req_q = ctx.Queue()
-            self.inf_sems = [ctx.Semaphore(0) for _ in range(args.num_actors)]
+            self.inf_sems = [ctx.Semaphore(0) for _ in range(rows)]
             self.weights_version = ctx.Value("l", 0)
             self.inf_pause = ctx.Value("i", 0)
             self.inf_paused_ack = ctx.Value("i", 0)
@@ -160,14 +170,26 @@ class ImpalaTrainer:
             self.inference_proc.start()
 
         for i in range(args.num_actors):
-            kw = dict(seed=args.seed, episode_queue=self.episode_q)
-            if self.inference == "gpu":
-                kw.update(inf_slots=self.inf_slots, inf_req_q=self.inf_req_q,
-                          inf_sem=self.inf_sems[i])
+            if self.double_buffer:
+                from ..parallel.rollout import actor_loop_db
+                target = actor_loop_db
+                kw = dict(seed=args.seed, episode_queue=self.episode_q,
+                          inf_slots=self.inf_slots,
+                          inf_req_q=self.inf_req_q,
+                          inf_sems=(self.inf_sems[2 * i],
+                                    self.inf_sems[2 * i + 1]),
+                          row_ids=(2 * i, 2 * i + 1))
             else:
-                kw.update(actor_model=self.actor_model)
+                target = actor_loop
+                kw = dict(seed=args.seed, episode_queue=self.episode_q)
+                if self.inference == "gpu":
+                    kw.update(inf_slots=self.inf_slots,
+                              inf_req_q=self.inf_req_q,
+                              inf_sem=self.inf_sems[i])
+                else:
+                    kw.update(actor_model=self.actor_model)
             p = ctx.Process(
-                target=actor_loop,
+                target=target,
                 args=(i, env_spec, self.store, self.free_q, self.full_q,
                       self.stop_event, self.step_counter),
                 kwargs=kw, daemon=True,
@@ -544,7 +566,8 @@ class ImpalaTrainer:
 
     def shutdown(self) -> None:
         self.stop_event.set()
-        for _ in self.actors:
+        pills = len(self.actors) * (2 if self.double_buffer else 1)
+        for _ in range(pills):
             self.free_q.put(None)  # poison pills (impala_atari.py:480)
         if self.inference == "gpu":
             for s in self.inf_sems:
