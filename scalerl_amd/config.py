@@ -179,6 +179,11 @@ class ImpalaArguments(RLArguments):
     output_dir: str = _h("checkpoint/log dir", default="work_dirs/impala")
     checkpoint_path: str = _h("explicit checkpoint file (empty → auto)", default="")
     inference: str = _h("actor inference placement: cpu | gpu", default="cpu")
+    inference_worker: str = _h(
+        "gpu-inference execution: 'thread' = a thread of the learner "
+        "process on its own HIP stream (one context — robust with "
+        "hipGraphs on ROCm 7.2); 'process' = dedicated process",
+        default="thread")
     remote_actor_slots: int = _h("store slots reserved for remote-node "
                                  "actors (0 disables the TCP server)", default=0)
     remote_port: int = _h("TCP port for remote actor nodes (0 = ephemeral)", default=0)

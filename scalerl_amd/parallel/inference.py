@@ -111,26 +111,39 @@ class RemotePolicy:
 def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlots,
                      shared_flat: torch.Tensor, version: "mp.Value",
                      req_q, sems: List, stop_event, max_batch_actors: int = 0,
-                     seed: int = 0, pause_flag=None, paused_ack=None):
-    """Inference worker process main."""
+                     seed: int = 0, pause_flag=None, paused_ack=None,
+                     as_thread: bool = False):
+    """Inference worker main — either a dedicated process (own HIP context)
+    or a THREAD of the learner process on its own stream (``as_thread``).
+
+    The thread mode exists because hipGraph execution in one process with
+    concurrent HIP submissions from another is fragile on ROCm 7.2 (r1/r2
+    measured HSA_STATUS_ERROR_EXCEPTION aborts both during capture and
+    during replay); a same-context second stream plus
+    ``capture_error_mode="thread_local"`` is the supported shape for this.
+    """
     from ..models.atari import AtariNet
     from .flat import FlatParams
     from .rollout import pin_tensor
 
     import os
-    torch.manual_seed(seed)
     device = torch.device(f"cuda:{device_index}")
-    torch.cuda.set_device(device)
+    if not as_thread:
+        torch.manual_seed(seed)
+        torch.cuda.set_device(device)
+    stream = torch.cuda.Stream(device=device) if as_thread else None
+    done_ev = torch.cuda.Event()
     model = AtariNet(**model_kwargs).to(device)
     model.train()  # multinomial action sampling
     flat = FlatParams(model, device=device)
     for t in slots.tensors():
         pin_tensor(t)
-    pin_tensor(shared_flat)
+    if not as_thread:  # the learner already pinned it in thread mode
+        pin_tensor(shared_flat)
     # bf16 convs/FC for the behavior forward: the recorded behavior logits
     # are the ones actions are sampled from, so the V-trace correction
     # stays exact; heads/LSTM run fp32 inside AtariNet regardless.
-    inf_bf16 = bool(os.environ.get("SCALERL_INF_BF16"))
+    inf_bf16 = os.environ.get("SCALERL_INF_BF16", "1") != "0"
 
     A, E = slots.num_actors, slots.envs_per_actor
     nact = slots.num_actions
@@ -149,13 +162,72 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
     done_d = torch.empty((A, E), dtype=torch.bool, device=device)
     lastact_d = torch.empty((A, E), dtype=torch.int64, device=device)
 
+    nonlocal_state = {"seen_version": seen_version}
+
+    def run_round(ids):
+        if version.value != nonlocal_state["seen_version"]:
+            nonlocal_state["seen_version"] = version.value
+            flat.load_from(shared_flat, non_blocking=False)
+
+        # FIXED-SHAPE round: forward ALL A actor slots every time (only
+        # requesting actors' inputs are refreshed and only their states/
+        # outputs are written back).  A varying batch size would retrigger
+        # MIOpen find per new shape and preclude graph capture.
+        for a in ids:
+            obs_d[a].copy_(slots.obs[a], non_blocking=True)
+            rew_d[a].copy_(slots.reward[a], non_blocking=True)
+            done_d[a].copy_(slots.done[a], non_blocking=True)
+            lastact_d[a].copy_(slots.last_action[a], non_blocking=True)
+        inputs = {
+            "obs": obs_d.reshape(1, A * E, *slots.obs.shape[2:]),
+            "reward": rew_d.reshape(1, A * E),
+            "done": done_d.reshape(1, A * E),
+            "last_action": lastact_d.reshape(1, A * E),
+        }
+        state = ()
+        snap = {}
+        if use_lstm:
+            for a in ids:
+                if slots.want_state[a]:
+                    snap[a] = (h_all[:, a * E:(a + 1) * E].clone(),
+                               c_all[:, a * E:(a + 1) * E].clone())
+            # no clone: the LSTM builds fresh state tensors, and only
+            # requesting actors' slices are copied back below — cloning
+            # h/c here cost ~50 MB of D2D per round at A=24,E=128
+            state = (h_all, c_all)
+        if inf_bf16:
+            with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+                out, new_state = model(inputs, state)
+        else:
+            out, new_state = model(inputs, state)
+        action = out["action"].view(A, E)
+        logits = out["policy_logits"].view(A, E, nact)
+        for a in ids:
+            if use_lstm:
+                sl = slice(a * E, (a + 1) * E)
+                h_all[:, sl] = new_state[0][:, sl]
+                c_all[:, sl] = new_state[1][:, sl]
+            slots.action[a].copy_(action[a], non_blocking=True)
+            slots.logits[a].copy_(logits[a], non_blocking=True)
+            if a in snap:
+                hj, cj = snap[a]
+                slots.core_state[a][0].copy_(hj, non_blocking=True)
+                slots.core_state[a][1].copy_(cj, non_blocking=True)
+        # fence only OUR stream before releasing the actors (a device-wide
+        # synchronize in thread mode would stall the learner stream too)
+        done_ev.record(stream if stream is not None
+                       else torch.cuda.current_stream(device))
+        done_ev.synchronize()
+
+    import contextlib
     with torch.no_grad():
         while not stop_event.is_set():
             if pause_flag is not None and pause_flag.value:
                 # learner is hipGraph-capturing on this device: quiesce our
                 # HIP queue entirely until it clears the flag (concurrent
                 # submissions from another process during stream capture
-                # fault the HSA queue on ROCm 7.2)
+                # fault the HSA queue on ROCm 7.2; thread mode makes this
+                # path unnecessary but it is kept for process mode)
                 torch.cuda.synchronize()
                 if paused_ack is not None:
                     paused_ack.value = 1
@@ -180,54 +252,9 @@ def inference_worker(device_index: int, model_kwargs: Dict, slots: InferenceSlot
                     stop_event.set()
                     break
                 ids.append(nxt)
-            if version.value != seen_version:
-                seen_version = version.value
-                flat.load_from(shared_flat, non_blocking=False)
-
-            # FIXED-SHAPE round: forward ALL A actor slots every time (only
-            # requesting actors' inputs are refreshed and only their states/
-            # outputs are written back).  A varying batch size would retrigger
-            # MIOpen find per new shape and preclude graph capture.
-            for a in ids:
-                obs_d[a].copy_(slots.obs[a], non_blocking=True)
-                rew_d[a].copy_(slots.reward[a], non_blocking=True)
-                done_d[a].copy_(slots.done[a], non_blocking=True)
-                lastact_d[a].copy_(slots.last_action[a], non_blocking=True)
-            inputs = {
-                "obs": obs_d.reshape(1, A * E, *slots.obs.shape[2:]),
-                "reward": rew_d.reshape(1, A * E),
-                "done": done_d.reshape(1, A * E),
-                "last_action": lastact_d.reshape(1, A * E),
-            }
-            state = ()
-            snap = {}
-            if use_lstm:
-                for a in ids:
-                    if slots.want_state[a]:
-                        snap[a] = (h_all[:, a * E:(a + 1) * E].clone(),
-                                   c_all[:, a * E:(a + 1) * E].clone())
-                # no clone: the LSTM builds fresh state tensors, and only
-                # requesting actors' slices are copied back below — cloning
-                # h/c here cost ~50 MB of D2D per round at A=24,E=128
-                state = (h_all, c_all)
-            if inf_bf16:
-                with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
-                    out, new_state = model(inputs, state)
-            else:
-                out, new_state = model(inputs, state)
-            action = out["action"].view(A, E)
-            logits = out["policy_logits"].view(A, E, nact)
-            for a in ids:
-                if use_lstm:
-                    sl = slice(a * E, (a + 1) * E)
-                    h_all[:, sl] = new_state[0][:, sl]
-                    c_all[:, sl] = new_state[1][:, sl]
-                slots.action[a].copy_(action[a], non_blocking=True)
-                slots.logits[a].copy_(logits[a], non_blocking=True)
-                if a in snap:
-                    hj, cj = snap[a]
-                    slots.core_state[a][0].copy_(hj, non_blocking=True)
-                    slots.core_state[a][1].copy_(cj, non_blocking=True)
-            torch.cuda.synchronize()
+            sctx = (torch.cuda.stream(stream) if stream is not None
+                    else contextlib.nullcontext())
+            with sctx:
+                run_round(ids)
             for a in ids:
                 sems[a].release()

@@ -159,15 +159,24 @@ class ImpalaTrainer:
                                 num_actions=self.num_actions,
                                 use_lstm=args.use_lstm)
             dev_index = self.device.index or 0
-            self.inference_proc = ctx.Process(
-                target=inference_worker,
-                args=(dev_index, model_kwargs, self.inf_slots,
-                      self.shared_flat.flat, self.weights_version,
-                      self.inf_req_q, self.inf_sems, self.stop_event),
-                kwargs=dict(seed=args.seed + 9999, pause_flag=self.inf_pause,
-                            paused_ack=self.inf_paused_ack),
-                daemon=True, name=f"impala-infer-{self.rank}")
-            self.inference_proc.start()
+            wargs = (dev_index, model_kwargs, self.inf_slots,
+                     self.shared_flat.flat, self.weights_version,
+                     self.inf_req_q, self.inf_sems, self.stop_event)
+            if getattr(args, "inference_worker", "thread") == "thread":
+                import threading
+                self.inference_proc = threading.Thread(
+                    target=inference_worker, args=wargs,
+                    kwargs=dict(seed=args.seed + 9999, as_thread=True),
+                    daemon=True, name=f"impala-infer-{self.rank}")
+                self.inference_proc.start()
+            else:
+                self.inference_proc = ctx.Process(
+                    target=inference_worker, args=wargs,
+                    kwargs=dict(seed=args.seed + 9999,
+                                pause_flag=self.inf_pause,
+                                paused_ack=self.inf_paused_ack),
+                    daemon=True, name=f"impala-infer-{self.rank}")
+                self.inference_proc.start()
 
         for i in range(args.num_actors):
             if self.double_buffer:
@@ -291,8 +300,11 @@ class ImpalaTrainer:
                           f"{self.remote_server.port}")
 
     def _pause_inference(self, pause: bool, timeout_s: float = 30.0) -> None:
-        """Quiesce the inference worker's HIP queue around graph capture."""
+        """Quiesce the inference worker's HIP queue around graph capture
+        (process mode only: thread-mode capture uses thread_local scope)."""
         if self.inference != "gpu":
+            return
+        if getattr(self.args, "inference_worker", "thread") == "thread":
             return
         if pause:
             self.inf_pause.value = 1
@@ -583,8 +595,10 @@ class ImpalaTrainer:
         self.actors.clear()
         if self.inference_proc is not None:
             self.inference_proc.join(timeout=2.0)
-            if self.inference_proc.is_alive():
-                self.inference_proc.terminate()
+            if (self.inference_proc.is_alive()
+                    and hasattr(self.inference_proc, "terminate")):
+                self.inference_proc.terminate()  # Process only; a Thread
+                # is daemon and dies with us
             self.inference_proc = None
         if getattr(self, "remote_server", None) is not None:
             self.remote_server.close()
