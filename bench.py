@@ -56,6 +56,8 @@ def parse_args():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--inference", type=str, default="auto",
                    choices=["auto", "cpu", "gpu"])
+    p.add_argument("--inference-worker", type=str, default="thread",
+                   choices=["thread", "process"])
     return p.parse_args()
 
 
@@ -118,6 +120,7 @@ def main():
         use_lstm=bool(args.use_lstm), device=device, dtype=args.dtype,
         use_graph=bool(args.use_graph),
         actor_double_buffer=bool(args.double_buffer),
+        inference_worker=args.inference_worker,
         inference=inference, seed=1234 + rank,
         total_steps=1 << 60, disable_checkpoint=True,
         output_dir="/tmp/scalerl_bench")

@@ -279,7 +279,14 @@ class MaskedLSTM(nn.Module):
         hs_out: List[torch.Tensor] = []
         cs_out: List[torch.Tensor] = []
         out = x
-        fn = (_MaskedLSTMSeqFn if (x.is_cuda and _use_seq_path())
+        # The C++ seq loop drives a process-global rocBLAS handle, which is
+        # NOT safe to call from two threads at once (the thread-mode
+        # inference worker shares the process with the learner).  T==1 has
+        # no launch-overhead to amortize anyway, so single-step forwards
+        # (the inference worker's shape) take the per-step path, keeping
+        # the seq loop learner-only.
+        fn = (_MaskedLSTMSeqFn if (x.is_cuda and x.shape[0] > 1 and
+                                   _use_seq_path())
               else _MaskedLSTMFn)
         for k in range(self.num_layers):
             out, hN, cN = fn.apply(
