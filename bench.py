@@ -47,7 +47,7 @@ def parse_args():
     p.add_argument("--double-buffer", type=int, default=0,
                    help="two env groups per actor (SEED latency hiding)")
     p.add_argument("--batch-size", type=int, default=256)
-    p.add_argument("--envs-per-actor", type=int, default=128)
+    p.add_argument("--envs-per-actor", type=int, default=256)
     p.add_argument("--num-actors", type=int, default=0,
                    help="actor procs per rank (0 = auto from cpu count)")
     p.add_argument("--use-lstm", type=int, default=1)
@@ -56,7 +56,7 @@ def parse_args():
     p.add_argument("--device", type=str, default="auto")
     p.add_argument("--inference", type=str, default="auto",
                    choices=["auto", "cpu", "gpu"])
-    p.add_argument("--inference-worker", type=str, default="thread",
+    p.add_argument("--inference-worker", type=str, default="process",
                    choices=["thread", "process"])
     return p.parse_args()
 
@@ -107,6 +107,12 @@ def main():
             slots = args.num_actors + 2 * max(args.batch_size // E, 1) + 2
             slot = (args.rollout_length + 1) * E * (4 * 84 * 84 + 6 * 4 + 28)
             return slots * slot
+        while (args.num_actors > 8
+               and world * store_bytes(args.envs_per_actor) > 0.8 * shm_free):
+            args.num_actors -= 4
+            if rank == 0:
+                print(f"[bench] /dev/shm pressure: num_actors -> "
+                      f"{args.num_actors}", flush=True)
         while (args.envs_per_actor > 16
                and world * store_bytes(args.envs_per_actor) > 0.8 * shm_free):
             args.envs_per_actor //= 2

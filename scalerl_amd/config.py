@@ -180,10 +180,10 @@ class ImpalaArguments(RLArguments):
     checkpoint_path: str = _h("explicit checkpoint file (empty → auto)", default="")
     inference: str = _h("actor inference placement: cpu | gpu", default="cpu")
     inference_worker: str = _h(
-        "gpu-inference execution: 'thread' = a thread of the learner "
-        "process on its own HIP stream (one context — robust with "
-        "hipGraphs on ROCm 7.2); 'process' = dedicated process",
-        default="thread")
+        "gpu-inference execution: 'process' = dedicated process (default; "
+        "a same-process 'thread' variant exists but aborts with HSA "
+        "exceptions on ROCm 7.2 — see profiles/README.md)",
+        default="process")
     remote_actor_slots: int = _h("store slots reserved for remote-node "
                                  "actors (0 disables the TCP server)", default=0)
     remote_port: int = _h("TCP port for remote actor nodes (0 = ephemeral)", default=0)

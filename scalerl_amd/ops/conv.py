@@ -76,11 +76,13 @@ def atari_conv_fwd(layer: int, x: torch.Tensor, weight: torch.Tensor,
     b = bias.float().contiguous() if bias is not None else None
     out = torch.empty((x.shape[0], *out_shape), dtype=torch.bfloat16,
                       device=x.device)
-    ret = fn(_c(x.contiguous().data_ptr()), _c(w.data_ptr()),
+    xc = x.contiguous()  # local outlives the launch (see atari_conv_wgrad)
+    ret = fn(_c(xc.data_ptr()), _c(w.data_ptr()),
              _c(b.data_ptr()) if b is not None else None,
              _c(out.data_ptr()), x.shape[0], int(relu),
              _backend.current_stream())
     _backend.check(ret, name)
+    del xc
     return out
 
 
@@ -112,7 +114,13 @@ _WSHAPE = {1: (32, 4, 8, 8), 2: (64, 32, 4, 4), 3: (64, 64, 3, 3)}
 @torch.no_grad()
 def atari_conv_wgrad(layer: int, x: torch.Tensor, dout: torch.Tensor,
                      split: int = 64) -> torch.Tensor:
-    """dL/dW for one encoder conv (fp32 out).  Layer 1 takes uint8 x."""
+    """dL/dW for one encoder conv (fp32 out).  Layer 1 takes uint8 x.
+
+    NOTE every cast/contiguous result is BOUND TO A LOCAL that outlives
+    the launch: taking ``.data_ptr()`` of an unreferenced temporary frees
+    its block before ``fn`` is even called, and the NEXT argument's
+    allocation can reuse and overwrite it (this exact use-after-free
+    corrupted dgrad results for weeks — r2 debug log, profiles/README.md)."""
     lib = _declare_bwd(_declare_conv(_backend.lib()))
     K, C, KH, KW = _WSHAPE[layer]
     dw = torch.zeros(K, C * KH * KW, device=x.device, dtype=torch.float32)
@@ -121,27 +129,31 @@ def atari_conv_wgrad(layer: int, x: torch.Tensor, dout: torch.Tensor,
         xc = x.contiguous()
     else:
         xc = x.to(torch.bfloat16).contiguous()
+    dc = dout.to(torch.bfloat16).contiguous()
     fn = getattr(lib, _WGRAD[layer])
-    ret = fn(_c(xc.data_ptr()),
-             _c(dout.to(torch.bfloat16).contiguous().data_ptr()),
+    ret = fn(_c(xc.data_ptr()), _c(dc.data_ptr()),
              _c(dw.data_ptr()), x.shape[0], split, _backend.current_stream())
     _backend.check(ret, _WGRAD[layer])
+    del xc, dc  # keep alive past the (async, same-stream) launch
     return dw.view(K, C, KH, KW)
 
 
 @torch.no_grad()
 def atari_conv_dgrad(layer: int, dout: torch.Tensor,
                      weight: torch.Tensor) -> torch.Tensor:
-    """dL/dX for encoder convs 2-3 (bf16 out; conv1 is the input layer)."""
+    """dL/dX for encoder convs 2-3 (bf16 out; conv1 is the input layer).
+    Temp-lifetime discipline: see atari_conv_wgrad."""
     lib = _declare_bwd(_declare_conv(_backend.lib()))
     _, in_shape, _ = _SHAPES[layer]
     din = torch.empty((dout.shape[0], *in_shape), dtype=torch.bfloat16,
                       device=dout.device)
+    dc = dout.to(torch.bfloat16).contiguous()
+    wc = weight.to(torch.bfloat16).contiguous()
     fn = getattr(lib, _DGRAD[layer])
-    ret = fn(_c(dout.to(torch.bfloat16).contiguous().data_ptr()),
-             _c(weight.to(torch.bfloat16).contiguous().data_ptr()),
+    ret = fn(_c(dc.data_ptr()), _c(wc.data_ptr()),
              _c(din.data_ptr()), dout.shape[0], _backend.current_stream())
     _backend.check(ret, _DGRAD[layer])
+    del dc, wc
     return din
 
 

@@ -53,11 +53,16 @@ class _PPOFusedLossFn(torch.autograd.Function):
         grad_logits = torch.empty_like(lg)
         grad_values = torch.empty_like(vals)
         loss_out = torch.zeros(3, device=lg.device)
+        # locals keep the cast temps alive past the launch (see ops/td.py)
+        act = actions.contiguous().long()
+        olp = old_logp.contiguous().float()
+        advc = adv.contiguous().float()
+        retc = returns.contiguous().float()
         ret = lib.ppo_fused_loss(
-            _c(lg.data_ptr()), _c(actions.contiguous().long().data_ptr()),
-            _c(old_logp.contiguous().float().data_ptr()),
-            _c(adv.contiguous().float().data_ptr()),
-            _c(returns.contiguous().float().data_ptr()),
+            _c(lg.data_ptr()), _c(act.data_ptr()),
+            _c(olp.data_ptr()),
+            _c(advc.data_ptr()),
+            _c(retc.data_ptr()),
             _c(vals.data_ptr()), clip_eps, vcoef, ecoef, N, A,
             _c(grad_logits.data_ptr()), _c(grad_values.data_ptr()),
             _c(loss_out.data_ptr()), _backend.current_stream())

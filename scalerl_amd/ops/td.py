@@ -56,13 +56,23 @@ class _FusedTDLossFn(torch.autograd.Function):
         grad_q = torch.zeros_like(qc)
         td_abs = torch.empty(B, device=q.device, dtype=torch.float32)
         loss_out = torch.zeros(1, device=q.device, dtype=torch.float32)
+        # bind every cast/contiguous temp to a local that outlives the
+        # launch: a data_ptr() taken from an unreferenced temp is freed
+        # before the ctypes call runs, and the next argument's allocation
+        # can reuse+overwrite the block (the r2 dgrad corruption)
+        qno = (q_next_online.contiguous().float()
+               if q_next_online is not None else None)
+        qnt = q_next_target.contiguous().float()
+        act = actions.contiguous().long()
+        rew = rewards.contiguous().float()
+        dis = discounts.contiguous().float()
         ret = _backend.lib().fused_td_loss(
             _c(qc.data_ptr()),
-            _c(q_next_online.contiguous().float().data_ptr()) if q_next_online is not None else None,
-            _c(q_next_target.contiguous().float().data_ptr()),
-            _c(actions.contiguous().long().data_ptr()),
-            _c(rewards.contiguous().float().data_ptr()),
-            _c(discounts.contiguous().float().data_ptr()),
+            _c(qno.data_ptr()) if qno is not None else None,
+            _c(qnt.data_ptr()),
+            _c(act.data_ptr()),
+            _c(rew.data_ptr()),
+            _c(dis.data_ptr()),
             _c(prios.data_ptr()) if prios is not None else None,
             _c(p_total.data_ptr()) if p_total is not None else None,
             _c(p_min.data_ptr()) if p_min is not None else None,

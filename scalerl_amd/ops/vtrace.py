@@ -114,12 +114,16 @@ class _ImpalaFusedLossFn(torch.autograd.Function):
         loss_out = torch.zeros(3, device=tl.device, dtype=torch.float32)
         vs_out = torch.empty_like(vals) if want_vs else None
         c = ctypes.c_void_p
+        # locals keep the cast temps alive past the launch (see ops/td.py)
+        rew = rewards.contiguous().float()
+        dis = discounts.contiguous().float()
+        bv = bootstrap_value.contiguous().float()
         ret = _backend.lib().impala_fused_loss(
             c(bl.data_ptr()), c(tl.data_ptr()), c(acts.data_ptr()),
-            c(rewards.contiguous().float().data_ptr()),
-            c(discounts.contiguous().float().data_ptr()),
+            c(rew.data_ptr()),
+            c(dis.data_ptr()),
             c(vals.data_ptr()),
-            c(bootstrap_value.contiguous().float().data_ptr()),
+            c(bv.data_ptr()),
             clip_rho, clip_c, clip_pg_rho, baseline_cost, entropy_cost,
             T, B, A, c(grad_logits.data_ptr()), c(grad_values.data_ptr()),
             c(loss_out.data_ptr()),

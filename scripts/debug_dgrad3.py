@@ -56,5 +56,33 @@ def main():
         print(f"layer{layer} with-sync+cloned-args: bad {int(bad.sum())}")
 
 
+def probe2():
+    """Decisive probes: (a) identical images — per-block outputs must be
+    identical; (b) batch sweep — where does the wrong/right boundary sit?"""
+    layer = 3
+    C, KH, KW, S, IH, IW, OH, OW, KOUT = CFG[layer]
+    torch.manual_seed(2)
+    w = torch.randn(KOUT, C, KH, KW).cuda() * 0.1
+    d1 = torch.randn(1, KOUT, OH, OW).cuda()
+    # (a) identical images
+    N = 12
+    dout = d1.repeat(N, 1, 1, 1)
+    got = atari_conv_dgrad(layer, dout, w).float()
+    ref0 = atari_conv_dgrad(layer, d1, w).float()[0]
+    per_img = [(got[i] - ref0).abs().max().item() for i in range(N)]
+    print(f"identical-images: max|got[i]-got_single| per image {per_img}")
+    # (b) batch sweep with distinct images
+    torch.manual_seed(3)
+    douts = torch.randn(13, KOUT, OH, OW).cuda()
+    singles = torch.stack([
+        atari_conv_dgrad(layer, douts[i:i + 1], w).float()[0]
+        for i in range(13)])
+    for B in (2, 4, 7, 13):
+        gotB = atari_conv_dgrad(layer, douts[:B], w).float()
+        diffs = [round((gotB[i] - singles[i]).abs().max().item(), 3)
+                 for i in range(B)]
+        print(f"batch={B}: max|batched-single| per image {diffs}")
+
+
 if __name__ == "__main__":
-    main()
+    probe2()
