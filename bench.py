@@ -42,8 +42,9 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--rollout-length", type=int, default=80)
-    p.add_argument("--use-graph", type=int, default=0,
-                   help="hipGraph-capture the learner step")
+    p.add_argument("--use-graph", type=int, default=1,
+                   help="hipGraph-capture the learner step (measured 539k "
+                        "vs 340k eager at the default config)")
     p.add_argument("--double-buffer", type=int, default=0,
                    help="two env groups per actor (SEED latency hiding)")
     p.add_argument("--batch-size", type=int, default=256)

@@ -244,7 +244,14 @@ class ImpalaTrainer:
         # start_actors() (callers: bench.py, train()).  The pause handshake
         # remains as a fallback for the lazy capture path.
         if self.use_graph and self.device.type == "cuda" and not self._started:
-            self._capture_graph()
+            try:
+                self._capture_graph()
+            except Exception as e:  # capture is an optimization, not a
+                # correctness requirement: fall back to the eager step
+                self.log.warning(f"hipGraph capture failed ({e!r}); "
+                                 f"falling back to the eager learner step")
+                self._graphed = None
+                self.use_graph = False
         self._setup_remote_server()
 
     def _capture_graph(self) -> None:

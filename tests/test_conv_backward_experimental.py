@@ -1,8 +1,8 @@
 """MFMA conv backward (wgrad/dgrad) vs torch autograd oracle.
 
-Gated behind SCALERL_EXPERIMENTAL=1 until hardware-validated (the forward
-kernels and fragment maps they share ARE validated — see
-test_conv_experimental.py)."""
+Hardware-validated r2 (10/10 on MI355X) after fixing the ctypes wrapper
+temp-lifetime bug that had been corrupting dgrad inputs
+(profiles/README.md r2 finding 3)."""
 
 import os
 
@@ -10,11 +10,7 @@ import pytest
 import torch
 import torch.nn.functional as F
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(not os.environ.get("SCALERL_EXPERIMENTAL"),
-                       reason="unvalidated backward kernels"),
-]
+pytestmark = [pytest.mark.gpu]
 
 SHAPES = {1: ((4, 84, 84), (32, 4, 8, 8), 4),
           2: ((32, 20, 20), (64, 32, 4, 4), 2),
