@@ -96,7 +96,8 @@ def _declare_bwd(lib):
         fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_long,
                        c.c_long, c.c_void_p]
         fn.restype = c.c_int
-    for name in ("atari_conv2_dgrad", "atari_conv3_dgrad"):
+    for name in ("atari_conv2_dgrad", "atari_conv3_dgrad",
+                 "atari_conv2_dgrad_v3"):
         fn = getattr(lib, name)
         fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_long,
                        c.c_void_p]
@@ -153,6 +154,26 @@ def atari_conv_dgrad(layer: int, dout: torch.Tensor,
     ret = fn(_c(dc.data_ptr()), _c(wc.data_ptr()),
              _c(din.data_ptr()), dout.shape[0], _backend.current_stream())
     _backend.check(ret, _DGRAD[layer])
+    del dc, wc
+    return din
+
+
+@torch.no_grad()
+def atari_conv2_dgrad_v3(dout: torch.Tensor,
+                         weight: torch.Tensor) -> torch.Tensor:
+    """Parity-decomposed stride-2 dgrad for conv2: 4x less MFMA work than
+    the masked v2 (csrc/conv_bwd.hip conv2_dgrad_v3).  EXPERIMENTAL until
+    hardware-validated (r3); temp-lifetime discipline as in
+    atari_conv_wgrad."""
+    lib = _declare_bwd(_declare_conv(_backend.lib()))
+    din = torch.empty((dout.shape[0], 32, 20, 20), dtype=torch.bfloat16,
+                      device=dout.device)
+    dc = dout.to(torch.bfloat16).contiguous()
+    wc = weight.to(torch.bfloat16).contiguous()
+    ret = lib.atari_conv2_dgrad_v3(
+        _c(dc.data_ptr()), _c(wc.data_ptr()), _c(din.data_ptr()),
+        dout.shape[0], _backend.current_stream())
+    _backend.check(ret, "atari_conv2_dgrad_v3")
     del dc, wc
     return din
 

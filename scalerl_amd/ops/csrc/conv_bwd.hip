@@ -261,3 +261,97 @@ extern "C" int atari_conv3_dgrad(const void* dout, const void* w, void* din,
   CHECK_LAUNCH();
   return 0;
 }
+
+// ------------------------------------------------ dgrad v3 (stride-2) --
+// Parity-decomposed input gradient for conv2 (4x4 stride 2): the masked
+// v2 formulation wastes 1/stride^2 = 4x of its MFMA work on taps whose
+// (iy-ky) % 2 != 0 (measured 27.9 ms = 45.8% of native-mode GPU time,
+// profiles/r2_micro_kernel_stats_native_conv.csv).  Here each of the 4
+// input-parity classes (py, px) reduces only over its OWN 2x2 sub-kernel:
+//   dx[n,c,2u+py,2v+px] = sum_{ko,a,b} dy[n,ko,u-a,v-b] * w[ko,c,py+2a,px+2b]
+// K = KOUT*4 = 256 (vs 1024 masked), boundary masks only.
+// EXPERIMENTAL until hardware-validated (SCALERL_EXPERIMENTAL gpu test).
+__global__ __launch_bounds__(256) void conv2_dgrad_v3(
+    const bf16_t* __restrict__ dout,    // [N, 64, 9, 9]
+    const bf16_t* __restrict__ weight,  // [64, 32, 4, 4]
+    bf16_t* __restrict__ dinput,        // [N, 32, 20, 20]
+    int batch) {
+  constexpr int C = 32, KOUT = 64, OH = 9, OW = 9, IH = 20, IW = 20;
+  constexpr int OPX = OH * OW, OPP = 88, IPX = IH * IW;
+  constexpr int HP = 10;                 // per-parity rows/cols (20/2)
+  constexpr int MP = HP * HP;            // 100 px per parity class
+  constexpr int KP = KOUT * 4;           // 256
+  constexpr int MT = (MP + 15) / 16;     // 7
+  constexpr int NT = C / 16;             // 2
+
+  __shared__ bf16_t dy[KOUT * OPP];      // 11264 B
+  __shared__ bf16_t dx[C * IPX];         // 25600 B
+
+  const int n = blockIdx.x;
+  for (int idx = threadIdx.x; idx < KOUT * OPX; idx += blockDim.x) {
+    const int ko = idx / OPX, px = idx % OPX;
+    dy[ko * OPP + px] = dout[((long)n * KOUT + ko) * OPX + px];
+  }
+  __syncthreads();
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g = lane >> 4, lr = lane & 15;
+
+  // 4 parity classes x 7 m-tiles x 2 n-tiles = 56 tiles
+  for (int t = wave; t < 4 * MT * NT; t += 4) {
+    const int par = t / (MT * NT);
+    const int rem = t % (MT * NT);
+    const int mt = rem / NT, nt = rem % NT;
+    const int py = par >> 1, px_par = par & 1;
+    const int row = mt * 16 + lr;        // parity-local pixel
+    const bool ok = row < MP;
+    const int u = row / HP, v = row % HP;
+    const int c = nt * 16 + lr;
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int kt = 0; kt < KP / 32; ++kt) {
+      const int k0 = kt * 32 + g * 8;
+      bf16x8 a, b;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int k = k0 + j;            // = ko*4 + a*2 + b
+        const int ko = k >> 2, ab = k & 3;
+        const int aa = ab >> 1, bb = ab & 1;
+        bf16_t av = (bf16_t)0.f;
+        if (ok) {
+          const int oy = u - aa, ox = v - bb;
+          if (oy >= 0 && oy < OH && ox >= 0 && ox < OW)
+            av = dy[ko * OPP + oy * OW + ox];
+        }
+        a[j] = av;
+        b[j] = weight[(((long)ko * C + c) * 4 + py + 2 * aa) * 4 +
+                      px_par + 2 * bb];
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int rw = mt * 16 + g * 4 + r;
+      if (rw < MP) {
+        const int uu = rw / HP, vv = rw % HP;
+        dx[(nt * 16 + lr) * IPX + (2 * uu + py) * IW + 2 * vv + px_par] =
+            (bf16_t)acc[r];
+      }
+    }
+  }
+  __syncthreads();
+
+  for (int idx = threadIdx.x; idx < C * IPX; idx += blockDim.x)
+    dinput[((long)n * C) * IPX + idx] = dx[idx];
+}
+
+extern "C" int atari_conv2_dgrad_v3(const void* dout, const void* w,
+                                    void* din, long batch,
+                                    hipStream_t stream) {
+  hipLaunchKernelGGL(conv2_dgrad_v3, dim3((unsigned)batch), dim3(256), 0,
+                     stream, (const bf16_t*)dout, (const bf16_t*)w,
+                     (bf16_t*)din, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}

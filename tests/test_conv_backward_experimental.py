@@ -60,3 +60,23 @@ def test_dgrad_matches_torch(layer):
     (out * dout).sum().backward()
     got = atari_conv_dgrad(layer, dout.cuda(), w.cuda()).float().cpu()
     torch.testing.assert_close(got, x.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.skipif(not os.environ.get("SCALERL_EXPERIMENTAL"),
+                    reason="dgrad v3 pending hardware validation (r3)")
+def test_dgrad_v3_matches_v2_and_torch():
+    """Stride-decomposed conv2 dgrad vs the validated v2 kernel and the
+    CPU oracle (math emulated in test_conv_formulas)."""
+    from scalerl_amd.ops.conv import atari_conv2_dgrad, atari_conv2_dgrad_v3
+    torch.manual_seed(6)
+    N = 13
+    x = torch.randn(N, 32, 20, 20, requires_grad=True)
+    w = torch.randn(64, 32, 4, 4) * 0.1
+    out = F.conv2d(x.to(torch.bfloat16).float(),
+                   w.to(torch.bfloat16).float(), stride=2)
+    dout = torch.randn_like(out)
+    (out * dout).sum().backward()
+    got3 = atari_conv2_dgrad_v3(dout.cuda(), w.cuda()).float().cpu()
+    got2 = atari_conv2_dgrad(2, dout.cuda(), w.cuda()).float().cpu()
+    torch.testing.assert_close(got3, x.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(got3, got2, rtol=2e-2, atol=2e-2)

@@ -230,3 +230,47 @@ def test_conv2_v2_fwd_formula_matches_torch():
         for ch in range(KOUT):
             out[n, ch].reshape(-1)[:] = np.maximum(out_lds[ch] + b[ch], 0.0)
     np.testing.assert_allclose(out, ref, rtol=2e-2, atol=2e-2)
+
+
+def test_conv2_dgrad_v3_parity_decomposition_matches_torch():
+    """Exact emulation of the stride-decomposed dgrad (conv_bwd.hip
+    conv2_dgrad_v3): each input-parity class reduces over its own 2x2
+    sub-kernel, K = KOUT*4 instead of the masked KOUT*16."""
+    torch.manual_seed(4)
+    N, C, KOUT, OH, OW, IH, IW, HP = 3, 32, 64, 9, 9, 20, 20, 10
+    x = torch.randn(N, C, IH, IW, requires_grad=True)
+    w = torch.randn(KOUT, C, 4, 4) * 0.1
+    out = F.conv2d(x.to(torch.bfloat16).float(),
+                   w.to(torch.bfloat16).float(), stride=2)
+    dout = torch.randn_like(out)
+    (out * dout).sum().backward()
+    want = x.grad.numpy()
+
+    dyb = dout.to(torch.bfloat16).float().numpy()
+    wb = w.to(torch.bfloat16).float().numpy()
+    got = np.zeros((N, C, IH, IW), dtype=np.float32)
+    for n in range(N):
+        for par in range(4):
+            py, px = par >> 1, par & 1
+            # A [MP, KP], B [KP, C] exactly as the kernel decodes
+            MP, KP = HP * HP, KOUT * 4
+            A = np.zeros((MP, KP), dtype=np.float32)
+            B = np.zeros((KP, C), dtype=np.float32)
+            for row in range(MP):
+                u, v = row // HP, row % HP
+                for k in range(KP):
+                    ko, ab = k >> 2, k & 3
+                    a, b = ab >> 1, ab & 1
+                    oy, ox = u - a, v - b
+                    if 0 <= oy < OH and 0 <= ox < OW:
+                        A[row, k] = dyb[n, ko, oy, ox]
+            for k in range(KP):
+                ko, ab = k >> 2, k & 3
+                a, b = ab >> 1, ab & 1
+                B[k, :] = wb[ko, :, py + 2 * a, px + 2 * b]
+            D = (torch.from_numpy(A).to(torch.bfloat16).float().numpy()
+                 @ torch.from_numpy(B).to(torch.bfloat16).float().numpy())
+            for row in range(MP):
+                u, v = row // HP, row % HP
+                got[n, :, 2 * u + py, 2 * v + px] = D[row, :]
+    np.testing.assert_allclose(got, want, rtol=5e-2, atol=5e-2)
