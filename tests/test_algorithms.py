@@ -167,3 +167,23 @@ def test_a3c_gpu_trainer_cpu_iteration():
         assert t.global_step == 3 * 8 * (2 * 4)
     finally:
         t.shutdown()
+
+
+@pytest.mark.gpu
+def test_a3c_gpu_trainer_on_gpu():
+    """Config-2 path on hardware: CPU actors (42×42 stack) + GPU learner
+    with GAE scan kernel and fused Adam."""
+    from scalerl_amd.config import A3CGpuArguments
+    from scalerl_amd.runtime.a3c_gpu import A3CGpuTrainer
+    args = A3CGpuArguments(num_actors=2, envs_per_actor=4, rollout_steps=8,
+                           slots_per_batch=2, device="cuda:0", seed=9,
+                           dtype="bf16", disable_checkpoint=True)
+    t = A3CGpuTrainer(args)
+    try:
+        t.setup_learner()
+        t.start_actors()
+        for _ in range(3):
+            s = t.train_iteration()
+            assert torch.isfinite(s["total_loss"])
+    finally:
+        t.shutdown()
