@@ -166,6 +166,9 @@ class A3CGpuTrainer:
                                  self.num_actions).to(self.device)
         self.model.load_state_dict(self.actor_model.state_dict())
         self.flat = FlatParams(self.model, device=self.device)
+        from ..parallel.dist import broadcast_flat, get_world_size as _ws
+        if _ws() > 1:  # dynamic: dist may init after __init__
+            broadcast_flat(self.flat.flat, src=0)  # identical init (DP)
         self.optimizer = FusedAdam(self.flat.flat, lr=args.learning_rate)
         self.gatherer = BatchGatherer(self.store, self.device,
                                       args.slots_per_batch)
@@ -245,8 +248,8 @@ class A3CGpuTrainer:
         self.timings.time("forward")
         total.backward()
         self.timings.time("backward")
-        if self.world_size > 1:
-            all_reduce_flat(self.flat.flat_grad, average=True)
+        all_reduce_flat(self.flat.flat_grad, average=True)  # no-op
+        # when torch.distributed is not initialized
         if args.max_grad_norm > 0:
             clip_grad_norm_(self.flat.flat_grad, args.max_grad_norm)
         self.optimizer.step(self.flat.flat_grad)

@@ -174,6 +174,10 @@ class ApexTrainer:
         self.target_model = self._build_qnet().to(dev)
         self.target_model.load_state_dict(self.model.state_dict())
         self.flat = FlatParams(self.model, device=dev)
+        if get_world_size() > 1:
+            from ..parallel import broadcast_flat
+            broadcast_flat(self.flat.flat, src=0)  # identical init (DP)
+            self.target_model.load_state_dict(self.model.state_dict())
         self.target_flat = FlatParams(self.target_model, device=dev)
         self.optimizer = FusedAdam(self.flat.flat, lr=args.learning_rate)
         if args.use_per:

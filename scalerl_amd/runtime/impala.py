@@ -217,6 +217,14 @@ class ImpalaTrainer:
         # identical init across ranks and with the actor model
         self.learner_model.load_state_dict(self.actor_model.state_dict())
         self.flat = FlatParams(self.learner_model, device=self.device)
+        # identical init on all ranks — per-rank seeds differ (they
+        # must, for decorrelated actors), so DP grad-averaging is only
+        # coherent after a rank-0 weight broadcast.  get_world_size() is
+        # read HERE, not in __init__: bench.py constructs the trainer
+        # before init_distributed().
+        from ..parallel.dist import broadcast_flat, get_world_size as _ws
+        if _ws() > 1:
+            broadcast_flat(self.flat.flat, src=0)
         self.optimizer = FusedRMSprop(
             self.flat.flat, lr=args.learning_rate, alpha=args.rmsprop_alpha,
             eps=args.rmsprop_eps, momentum=args.rmsprop_momentum)

@@ -25,6 +25,21 @@ def _dist_worker(rank, world, fn_name, port, q):
             t = torch.full((10,), float(rank * 7))
             broadcast_flat(t, src=0)
             q.put((rank, t[0].item()))  # everyone gets rank0's 0.0
+        elif fn_name == "a3cgpu":
+            from scalerl_amd.config import A3CGpuArguments
+            from scalerl_amd.runtime.a3c_gpu import A3CGpuTrainer
+            args = A3CGpuArguments(num_actors=1, envs_per_actor=4,
+                                   rollout_steps=6, slots_per_batch=1,
+                                   device="cpu", seed=20 + rank,
+                                   disable_checkpoint=True)
+            tr = A3CGpuTrainer(args)
+            tr.start_actors()
+            tr.setup_learner()
+            try:
+                tr.train_iteration()
+                q.put((rank, float(tr.flat.flat.sum())))
+            finally:
+                tr.shutdown()
         elif fn_name == "ddppo":
             from scalerl_amd.config import DDPPOArguments
             from scalerl_amd.runtime.ppo import DDPPOTrainer
@@ -72,3 +87,10 @@ def test_broadcast_flat_gloo():
 def test_ddppo_two_ranks_stay_in_sync():
     out = _run("ddppo", 29613)
     assert out[0] == pytest.approx(out[1], rel=1e-6)
+
+
+def test_a3c_gpu_trainer_two_ranks_stay_in_sync():
+    """A3CGpuTrainer's flat-grad all-reduce path keeps 2 CPU ranks
+    bit-identical after a synchronized update (config-2 DP over gloo)."""
+    out = _run("a3cgpu", 29731)
+    assert out[0] == pytest.approx(out[1], rel=0, abs=0)
