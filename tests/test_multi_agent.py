@@ -1,5 +1,6 @@
 """Multi-agent env layer + episode generator."""
 
+import pytest
 import numpy as np
 import torch
 
@@ -51,3 +52,28 @@ def test_episode_generator_roundtrip():
     r = [s["rewards"]["player_0"] for s in steps]
     R = ep["returns"]["player_0"]
     assert abs(R[0] - (r[0] + 0.5 * R[1])) < 1e-5
+
+
+def _have_pettingzoo() -> bool:
+    try:
+        import pettingzoo  # noqa: F401
+        return True
+    except ImportError:
+        return False
+
+
+@pytest.mark.skipif(not _have_pettingzoo(), reason="pettingzoo not in this "
+                    "image (no network); runs where it is installed")
+def test_real_pettingzoo_adapter_smoke():
+    """PettingZooAdapter over a real parallel PZ env (the in-image tests
+    cover the same API against SyntheticMultiAgentEnv fakes)."""
+    from pettingzoo.butterfly import pistonball_v6
+    from scalerl_amd.envs.multi_agent import PettingZooAdapter
+    env = PettingZooAdapter(pistonball_v6.parallel_env())
+    obs, _ = env.reset(seed=0)
+    assert set(obs) == set(env.agents)
+    for _ in range(5):
+        actions = {a: env.action_space(a).sample() for a in env.agents}
+        obs, rew, term, trunc, _ = env.step(actions)
+        assert set(rew) <= set(env.agents) | set(obs)
+    env.close()
