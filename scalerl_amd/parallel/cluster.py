@@ -197,6 +197,13 @@ class WorkerServer:
         # default bind is loopback; cross-node runs opt in with
         # bind="0.0.0.0" (or SCALERL_CLUSTER_BIND) + a shared secret
         bind = bind or os.environ.get("SCALERL_CLUSTER_BIND", "127.0.0.1")
+        if bind not in ("127.0.0.1", "localhost", "::1") and not self._secret:
+            import warnings
+            warnings.warn(
+                "cluster server binding a non-loopback interface with an "
+                "EMPTY shared secret — any host that can reach the port "
+                "can pull weights and push rollouts; set "
+                "SCALERL_CLUSTER_SECRET on every node", stacklevel=2)
         self._srv = socket.create_server((bind, port), backlog=64)
         self._srv.settimeout(0.5)
         self.port = self._srv.getsockname()[1]
