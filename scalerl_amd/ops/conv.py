@@ -97,7 +97,8 @@ def _declare_bwd(lib):
                        c.c_long, c.c_void_p]
         fn.restype = c.c_int
     for name in ("atari_conv2_dgrad", "atari_conv3_dgrad",
-                 "atari_conv2_dgrad_v3"):
+                 "atari_conv2_dgrad_v3", "atari_conv1_wgrad_v3",
+                 "atari_conv2_wgrad_v3", "atari_conv3_wgrad_v3"):
         fn = getattr(lib, name)
         fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_long,
                        c.c_void_p]
@@ -176,6 +177,34 @@ def atari_conv2_dgrad_v3(dout: torch.Tensor,
     _backend.check(ret, "atari_conv2_dgrad_v3")
     del dc, wc
     return din
+
+
+_WGRAD_V3 = {1: "atari_conv1_wgrad_v3", 2: "atari_conv2_wgrad_v3",
+             3: "atari_conv3_wgrad_v3"}
+
+
+@torch.no_grad()
+def atari_conv_wgrad_v3(layer: int, x: torch.Tensor,
+                        dout: torch.Tensor) -> torch.Tensor:
+    """Panel-staged wgrad (conv_bwd.hip convN_wgrad_v3): both MFMA
+    operands are contiguous LDS vector reads (v2's B side re-gathered
+    im2col scalar-by-scalar per tile).  EXPERIMENTAL until
+    hardware-validated (r3)."""
+    lib = _declare_bwd(_declare_conv(_backend.lib()))
+    K, C, KH, KW = _WSHAPE[layer]
+    dw = torch.zeros(K, C * KH * KW, device=x.device, dtype=torch.float32)
+    if layer == 1:
+        assert x.dtype == torch.uint8
+        xc = x.contiguous()
+    else:
+        xc = x.to(torch.bfloat16).contiguous()
+    dc = dout.to(torch.bfloat16).contiguous()
+    fn = getattr(lib, _WGRAD_V3[layer])
+    ret = fn(_c(xc.data_ptr()), _c(dc.data_ptr()), _c(dw.data_ptr()),
+             x.shape[0], _backend.current_stream())
+    _backend.check(ret, _WGRAD_V3[layer])
+    del xc, dc
+    return dw.view(K, C, KH, KW)
 
 
 class _NativeConvFn(torch.autograd.Function):

@@ -355,3 +355,130 @@ extern "C" int atari_conv2_dgrad_v3(const void* dout, const void* w,
   CHECK_LAUNCH();
   return 0;
 }
+
+// ------------------------------------------------ wgrad v3 (panel) -----
+// The v2 wgrad's B side re-gathers im2col values scalar-by-scalar for
+// every tile x ktile (r2c4 A/B: conv1 4.07 vs MIOpen ~1.24 ms).  v3
+// builds the quarter's im2col panel [KQ x PXC] in LDS ONCE per
+// (image, pixel-chunk), making BOTH MFMA operands contiguous vector
+// reads.  EXPERIMENTAL until hardware-validated (r3).
+//   PCH: pixel chunks per image (keeps LDS under the 64 KB static cap)
+//   PXC: padded chunk length (multiple of 32; zero-filled tail)
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT, int CQ, int NTQ, int TPW, int PCH, int PXC,
+          bool IN_U8 = false, typename in_t = bf16_t>
+__global__ __launch_bounds__(256) void convN_wgrad_v3(
+    const in_t* __restrict__ input, const bf16_t* __restrict__ dout,
+    float* __restrict__ dweight, int batch) {
+  constexpr int KDIM = C * KH * KW;
+  constexpr int KQ = CQ * KH * KW;
+  constexpr int MPX = OH * OW;
+  constexpr int CHUNK = (MPX + PCH - 1) / PCH;   // real px per chunk
+  constexpr int MT = KOUT / 16;
+
+  __shared__ bf16_t panel[KQ * PXC];
+  __shared__ bf16_t dy[KOUT * PXC];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g = lane >> 4, lr = lane & 15;
+  const int q = blockIdx.y;
+  const int c0 = q * CQ;
+
+  f32x4 acc[TPW];
+  #pragma unroll
+  for (int i = 0; i < TPW; ++i) acc[i] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int units = batch * PCH;
+  for (int u = blockIdx.x; u < units; u += gridDim.x) {
+    const int n = u / PCH, ch = u % PCH;
+    const int p0g = ch * CHUNK;
+    const int nreal = min(CHUNK, MPX - p0g);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < KOUT * PXC; idx += blockDim.x) {
+      const int ko = idx / PXC, p = idx % PXC;
+      dy[idx] = (p < nreal)
+          ? dout[((long)n * KOUT + ko) * MPX + p0g + p] : (bf16_t)0.f;
+    }
+    for (int idx = threadIdx.x; idx < KQ * PXC; idx += blockDim.x) {
+      const int kq = idx / PXC, p = idx % PXC;
+      bf16_t v = (bf16_t)0.f;
+      if (p < nreal) {
+        const int c = kq / (KH * KW);
+        const int r = kq % (KH * KW);
+        const int ky = r / KW, kx = r % KW;
+        const int pg = p0g + p;
+        const int oy = pg / OW, ox = pg % OW;
+        const in_t raw = input[(((long)n * C + c0 + c) * IH +
+                                oy * STRIDE + ky) * IW + ox * STRIDE + kx];
+        v = (bf16_t)(IN_U8 ? (float)raw * (1.0f / 255.0f) : (float)raw);
+      }
+      panel[idx] = v;
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ti = 0; ti < TPW; ++ti) {
+      const int t = wave + ti * 4;
+      const int mt = t / NTQ, nt = t % NTQ;
+      const int ko = mt * 16 + lr;
+      const int kq = nt * 16 + lr;
+      #pragma unroll 2
+      for (int pt = 0; pt < PXC / 32; ++pt) {
+        const int p0 = pt * 32 + g * 8;
+        const bf16x8 a = *(const bf16x8*)&dy[ko * PXC + p0];
+        const bf16x8 b = *(const bf16x8*)&panel[kq * PXC + p0];
+        acc[ti] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ti],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  #pragma unroll
+  for (int ti = 0; ti < TPW; ++ti) {
+    const int t = wave + ti * 4;
+    const int mt = t / NTQ, nt = t % NTQ;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int col = q * KQ + nt * 16 + lr;
+      const int row = mt * 16 + g * 4 + r;
+      atomicAdd(&dweight[(long)row * KDIM + col], acc[ti][r]);
+    }
+  }
+}
+
+extern "C" int atari_conv1_wgrad_v3(const void* in, const void* dout,
+                                    float* dw, long batch,
+                                    hipStream_t stream) {
+  // 2 chunks of 200 px (PXC 224); quarter = 1 channel; 2mt x 4nt, TPW 2
+  hipLaunchKernelGGL(
+      (convN_wgrad_v3<4, 8, 8, 4, 84, 84, 20, 20, 32, 1, 4, 2, 2, 224,
+                      true, unsigned char>),
+      dim3(wgrad_grid(2 * batch), 4), dim3(256), 0, stream,
+      (const unsigned char*)in, (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv2_wgrad_v3(const void* in, const void* dout,
+                                    float* dw, long batch,
+                                    hipStream_t stream) {
+  // 1 chunk of 81 px (PXC 96); quarter = 4 ch; 4mt x 4nt, TPW 4
+  hipLaunchKernelGGL(
+      (convN_wgrad_v3<32, 4, 4, 2, 20, 20, 9, 9, 64, 4, 4, 4, 1, 96>),
+      dim3(wgrad_grid(batch), 8), dim3(256), 0, stream, (const bf16_t*)in,
+      (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv3_wgrad_v3(const void* in, const void* dout,
+                                    float* dw, long batch,
+                                    hipStream_t stream) {
+  // 1 chunk of 49 px (PXC 64); quarter = 16 ch; 4mt x 9nt, TPW 9
+  hipLaunchKernelGGL(
+      (convN_wgrad_v3<64, 3, 3, 1, 9, 9, 7, 7, 64, 16, 9, 9, 1, 64>),
+      dim3(wgrad_grid(batch), 4), dim3(256), 0, stream, (const bf16_t*)in,
+      (const bf16_t*)dout, dw, (int)batch);
+  CHECK_LAUNCH();
+  return 0;
+}

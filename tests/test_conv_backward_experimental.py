@@ -80,3 +80,30 @@ def test_dgrad_v3_matches_v2_and_torch():
     got2 = atari_conv2_dgrad(2, dout.cuda(), w.cuda()).float().cpu()
     torch.testing.assert_close(got3, x.grad, rtol=5e-2, atol=5e-2)
     torch.testing.assert_close(got3, got2, rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.skipif(not os.environ.get("SCALERL_EXPERIMENTAL"),
+                    reason="wgrad v3 pending hardware validation (r3)")
+@pytest.mark.parametrize("layer", [1, 2, 3])
+def test_wgrad_v3_matches_v2_and_torch(layer):
+    """Panel-staged wgrad vs the validated v2 kernel and the CPU oracle."""
+    from scalerl_amd.ops.conv import atari_conv_wgrad, atari_conv_wgrad_v3
+    torch.manual_seed(8)
+    in_shape, w_shape, stride = SHAPES[layer]
+    N = 21
+    if layer == 1:
+        x_u8 = torch.randint(0, 256, (N, *in_shape), dtype=torch.uint8)
+        x_ref = (x_u8.float() / 255.0).to(torch.bfloat16).float()
+        x_in = x_u8.cuda()
+    else:
+        x = torch.randn(N, *in_shape)
+        x_ref = x.to(torch.bfloat16).float()
+        x_in = x.cuda()
+    w = (torch.randn(w_shape) * 0.1).requires_grad_()
+    dout = torch.randn(F.conv2d(x_ref, w, stride=stride).shape)
+    out = F.conv2d(x_ref, w.to(torch.bfloat16).float(), stride=stride)
+    (out * dout).sum().backward()
+    got3 = atari_conv_wgrad_v3(layer, x_in, dout.cuda()).cpu()
+    got2 = atari_conv_wgrad(layer, x_in, dout.cuda()).cpu()
+    torch.testing.assert_close(got3, w.grad, rtol=5e-2, atol=5e-1)
+    torch.testing.assert_close(got3, got2, rtol=2e-2, atol=2e-1)

@@ -274,3 +274,52 @@ def test_conv2_dgrad_v3_parity_decomposition_matches_torch():
                 u, v = row // HP, row % HP
                 got[n, :, 2 * u + py, 2 * v + px] = D[row, :]
     np.testing.assert_allclose(got, want, rtol=5e-2, atol=5e-2)
+
+
+def test_wgrad_v3_panel_decomposition_matches_torch():
+    """Exact emulation of the panel-staged wgrad (conv_bwd.hip
+    convN_wgrad_v3) for the conv1 instantiation (2 pixel chunks, PXC=224,
+    1-channel quarters): panel build + chunked-K GEMM accumulation."""
+    torch.manual_seed(7)
+    N, C, KH, KW, S, IH, IW, OH, OW, KOUT = 2, 4, 8, 8, 4, 84, 84, 20, 20, 32
+    CQ, PCH, PXC = 1, 2, 224
+    KQ = CQ * KH * KW
+    MPX = OH * OW
+    CHUNK = (MPX + PCH - 1) // PCH
+    x = torch.randint(0, 256, (N, C, IH, IW), dtype=torch.uint8)
+    w = (torch.randn(KOUT, C, KH, KW) * 0.1).requires_grad_()
+    x_ref = (x.float() / 255.0).to(torch.bfloat16).float()
+    out = F.conv2d(x_ref, w.to(torch.bfloat16).float(), stride=S)
+    dout = torch.randn_like(out)
+    (out * dout).sum().backward()
+    want = w.grad.numpy()
+
+    dyb = dout.to(torch.bfloat16).float().numpy()
+    xb = x_ref.numpy()
+    got = np.zeros((KOUT, C * KH * KW), dtype=np.float32)
+    for q in range(C // CQ):
+        c0 = q * CQ
+        acc = np.zeros((KOUT, KQ), dtype=np.float32)
+        for n in range(N):
+            for ch in range(PCH):
+                p0g = ch * CHUNK
+                nreal = min(CHUNK, MPX - p0g)
+                dy = np.zeros((KOUT, PXC), dtype=np.float32)
+                dy[:, :nreal] = dyb[n].reshape(KOUT, MPX)[:, p0g:p0g + nreal]
+                panel = np.zeros((KQ, PXC), dtype=np.float32)
+                for kq in range(KQ):
+                    c = kq // (KH * KW)
+                    r = kq % (KH * KW)
+                    ky, kx = r // KW, r % KW
+                    for p in range(nreal):
+                        pg = p0g + p
+                        oy, ox = pg // OW, pg % OW
+                        panel[kq, p] = xb[n, c0 + c, oy * S + ky,
+                                          ox * S + kx]
+                acc += (torch.from_numpy(dy).to(torch.bfloat16).float()
+                        .numpy() @
+                        torch.from_numpy(panel.T).to(torch.bfloat16)
+                        .float().numpy())
+        got[:, q * KQ:(q + 1) * KQ] = acc
+    np.testing.assert_allclose(got.reshape(KOUT, C, KH, KW), want,
+                               rtol=5e-2, atol=5e-1)
