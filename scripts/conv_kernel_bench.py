@@ -10,8 +10,9 @@ os.environ.setdefault("MIOPEN_FIND_MODE", "1")
 import torch
 import torch.nn.functional as F
 
-from scalerl_amd.ops.conv import (atari_conv_dgrad, atari_conv_fwd,
-                                  atari_conv_wgrad)
+from scalerl_amd.ops.conv import (atari_conv2_dgrad_v3, atari_conv_dgrad,
+                                  atari_conv_fwd, atari_conv_wgrad,
+                                  atari_conv_wgrad_v3)
 
 N = int(os.environ.get("CONV_BENCH_N", "20736"))
 dev = "cuda"
@@ -56,6 +57,8 @@ def main():
         print(f"conv{layer} fwd : native {t_nat:7.2f} ms  miopen {t_mio:7.2f} ms")
 
         t_nat = timeit(lambda: atari_conv_wgrad(layer, x, dout))
+        t_v3 = timeit(lambda: atari_conv_wgrad_v3(layer, x, dout))
+        print(f"conv{layer} wgrd v3: {t_v3:7.2f} ms (panel-staged)")
         wg = wb.clone().requires_grad_()
         def mio_wgrad():
             out = F.conv2d(x_t, wg, stride=stride)
@@ -66,6 +69,9 @@ def main():
 
         if layer in (2, 3):
             t_nat = timeit(lambda: atari_conv_dgrad(layer, dout, w))
+            if layer == 2:
+                t_v3 = timeit(lambda: atari_conv2_dgrad_v3(dout, w))
+                print(f"conv2 dgrd v3: {t_v3:7.2f} ms (stride-decomposed)")
             xg = x_t.clone().requires_grad_()
             def mio_dgrad():
                 out = F.conv2d(xg, wb.detach(), stride=stride)
