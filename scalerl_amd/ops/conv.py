@@ -23,7 +23,8 @@ def _declare_conv(lib):
         return lib
     c = ctypes
     for name in ("atari_conv1_fwd_u8", "atari_conv1_fwd_bf16",
-                 "atari_conv2_fwd", "atari_conv3_fwd"):
+                 "atari_conv2_fwd", "atari_conv3_fwd",
+                 "atari_conv2_fwd_v3", "atari_conv3_fwd_v3"):
         fn = getattr(lib, name)
         fn.argtypes = [c.c_void_p, c.c_void_p, c.c_void_p, c.c_void_p,
                        c.c_long, c.c_int, c.c_void_p]
@@ -181,6 +182,31 @@ def atari_conv2_dgrad_v3(dout: torch.Tensor,
 
 _WGRAD_V3 = {1: "atari_conv1_wgrad_v3", 2: "atari_conv2_wgrad_v3",
              3: "atari_conv3_wgrad_v3"}
+
+
+@torch.no_grad()
+def atari_conv_fwd_v3(layer: int, x: torch.Tensor, weight: torch.Tensor,
+                      bias: torch.Tensor = None,
+                      relu: bool = True) -> torch.Tensor:
+    """Panel-staged forward for conv2/conv3 (conv_fwd.hip convN_fwd_v3).
+    EXPERIMENTAL until hardware-validated (r3)."""
+    assert layer in (2, 3)
+    name, in_shape, out_shape = _SHAPES[layer]
+    assert tuple(x.shape[1:]) == in_shape
+    lib = _declare_conv(_backend.lib())
+    fn = getattr(lib, f"atari_conv{layer}_fwd_v3")
+    xc = x.to(torch.bfloat16).contiguous()
+    w = weight.to(torch.bfloat16).contiguous()
+    b = bias.float().contiguous() if bias is not None else None
+    out = torch.empty((x.shape[0], *out_shape), dtype=torch.bfloat16,
+                      device=x.device)
+    ret = fn(_c(xc.data_ptr()), _c(w.data_ptr()),
+             _c(b.data_ptr()) if b is not None else None,
+             _c(out.data_ptr()), x.shape[0], int(relu),
+             _backend.current_stream())
+    _backend.check(ret, f"atari_conv{layer}_fwd_v3")
+    del xc, w, b
+    return out
 
 
 @torch.no_grad()

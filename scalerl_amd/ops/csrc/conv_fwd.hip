@@ -312,3 +312,116 @@ extern "C" int atari_conv3_fwd(const void* in, const void* w,
   CHECK_LAUNCH();
   return 0;
 }
+
+// ------------------------------------------------- fwd v3 (panel) ------
+// conv2/conv3 forward with the im2col panel staged in LDS (K-quartered to
+// fit the 64 KB static cap), so the A side becomes contiguous vector
+// reads like the B side (v2's conv2 A-loads are 4x ds_read_b32 due to
+// ox-parity misalignment; conv3's are scalar).  conv1 fwd already beats
+// MIOpen in v2 form and is left alone.  EXPERIMENTAL until
+// hardware-validated (r3).
+//   NQ: K quarters; KQP = KDIM/NQ (multiple of 32)
+//   PADM: M rows padded to a multiple of 16 (garbage rows masked at store)
+template <int C, int KH, int KW, int STRIDE, int IH, int IW, int OH, int OW,
+          int KOUT, int NQ, int TPW>
+__global__ __launch_bounds__(256) void convN_fwd_v3(
+    const bf16_t* __restrict__ input,   // [N, C, IH, IW]
+    const bf16_t* __restrict__ weight,  // [KOUT, KDIM]
+    const float* __restrict__ bias,     // [KOUT] or nullptr
+    bf16_t* __restrict__ output,        // [N, KOUT, OH, OW]
+    int batch, int relu) {
+  constexpr int KDIM = C * KH * KW;
+  constexpr int KQP = KDIM / NQ;
+  constexpr int MPX = OH * OW;
+  constexpr int MT = (MPX + 15) / 16;
+  constexpr int PADM = MT * 16;
+  constexpr int NT = KOUT / 16;
+
+  __shared__ bf16_t panel[PADM * KQP];
+  __shared__ bf16_t out_lds[KOUT * MPX];
+
+  const int n = blockIdx.x;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g = lane >> 4, lr = lane & 15;
+
+  f32x4 acc[TPW];
+  #pragma unroll
+  for (int i = 0; i < TPW; ++i) acc[i] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  for (int qq = 0; qq < NQ; ++qq) {
+    __syncthreads();  // previous quarter's readers done
+    for (int idx = threadIdx.x; idx < MPX * KQP; idx += blockDim.x) {
+      const int px = idx / KQP, kq = idx % KQP;
+      const int k = qq * KQP + kq;
+      const int c = k / (KH * KW);
+      const int r = k % (KH * KW);
+      const int ky = r / KW, kx = r % KW;
+      const int oy = px / OW, ox = px % OW;
+      panel[px * KQP + kq] =
+          input[(((long)n * C + c) * IH + oy * STRIDE + ky) * IW +
+                ox * STRIDE + kx];
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int ti = 0; ti < TPW; ++ti) {
+      const int t = wave + ti * 4;
+      const int mt = t / NT, nt = t % NT;
+      const int px = mt * 16 + lr;              // may be a padded row:
+      const int ch = nt * 16 + lr;              // garbage masked at store
+      #pragma unroll 4
+      for (int kt = 0; kt < KQP / 32; ++kt) {
+        const int k0 = kt * 32 + g * 8;
+        const bf16x8 a = (px < MPX)
+            ? *(const bf16x8*)&panel[px * KQP + k0] : (bf16x8)(bf16_t)0.f;
+        const bf16x8 b =
+            *(const bf16x8*)&weight[(long)ch * KDIM + qq * KQP + k0];
+        acc[ti] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ti],
+                                                          0, 0, 0);
+      }
+    }
+  }
+  #pragma unroll
+  for (int ti = 0; ti < TPW; ++ti) {
+    const int t = wave + ti * 4;
+    const int mt = t / NT, nt = t % NT;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = mt * 16 + g * 4 + r;
+      if (row < MPX)
+        out_lds[(nt * 16 + lr) * MPX + row] = (bf16_t)acc[ti][r];
+    }
+  }
+  __syncthreads();
+  for (int idx = threadIdx.x; idx < KOUT * MPX; idx += blockDim.x) {
+    const int ch = idx / MPX;
+    float v = (float)out_lds[idx] + (bias ? bias[ch] : 0.f);
+    if (relu) v = fmaxf(v, 0.f);
+    output[((long)n * KOUT) * MPX + idx] = (bf16_t)v;
+  }
+}
+
+extern "C" int atari_conv2_fwd_v3(const void* in, const void* w,
+                                  const float* bias, void* out, long batch,
+                                  int relu, hipStream_t stream) {
+  // 6 mt x 4 nt = 24 tiles -> TPW 6; 2 K-quarters (panel 96x256 = 49 KB)
+  hipLaunchKernelGGL((convN_fwd_v3<32, 4, 4, 2, 20, 20, 9, 9, 64, 2, 6>),
+                     dim3((unsigned)batch), dim3(256), 0, stream,
+                     (const bf16_t*)in, (const bf16_t*)w, bias,
+                     (bf16_t*)out, (int)batch, relu);
+  CHECK_LAUNCH();
+  return 0;
+}
+
+extern "C" int atari_conv3_fwd_v3(const void* in, const void* w,
+                                  const float* bias, void* out, long batch,
+                                  int relu, hipStream_t stream) {
+  // 4 mt x 4 nt = 16 tiles -> TPW 4; 2 K-quarters (panel 64x288 = 36.9 KB)
+  hipLaunchKernelGGL((convN_fwd_v3<64, 3, 3, 1, 9, 9, 7, 7, 64, 2, 4>),
+                     dim3((unsigned)batch), dim3(256), 0, stream,
+                     (const bf16_t*)in, (const bf16_t*)w, bias,
+                     (bf16_t*)out, (int)batch, relu);
+  CHECK_LAUNCH();
+  return 0;
+}

@@ -11,8 +11,8 @@ import torch
 import torch.nn.functional as F
 
 from scalerl_amd.ops.conv import (atari_conv2_dgrad_v3, atari_conv_dgrad,
-                                  atari_conv_fwd, atari_conv_wgrad,
-                                  atari_conv_wgrad_v3)
+                                  atari_conv_fwd, atari_conv_fwd_v3,
+                                  atari_conv_wgrad, atari_conv_wgrad_v3)
 
 N = int(os.environ.get("CONV_BENCH_N", "20736"))
 dev = "cuda"
@@ -52,6 +52,9 @@ def main():
         doutf = dout.float()
 
         t_nat = timeit(lambda: atari_conv_fwd(layer, x, w, b))
+        if layer in (2, 3):
+            t_f3 = timeit(lambda: atari_conv_fwd_v3(layer, x, w, b))
+            print(f"conv{layer} fwd v3: {t_f3:7.2f} ms (panel-staged)")
         t_mio = timeit(lambda: F.relu(F.conv2d(x_t, wb, b.to(torch.bfloat16),
                                                stride=stride)))
         print(f"conv{layer} fwd : native {t_nat:7.2f} ms  miopen {t_mio:7.2f} ms")
