@@ -36,7 +36,7 @@ from ..data import PrioritizedReplayBuffer
 from ..models.atari import AtariQNet
 from ..ops import (FusedAdam, clip_grad_norm_, fused_polyak_, fused_td_loss,
                    nstep_fold)
-from ..parallel import FlatParams, all_reduce_flat, get_rank
+from ..parallel import FlatParams, all_reduce_flat, get_rank, get_world_size
 from ..parallel.rollout import RolloutStore, build_actor_env
 from ..utils import get_logger
 from ..utils.checkpoint import save_agent_checkpoint
