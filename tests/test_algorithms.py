@@ -49,6 +49,7 @@ def test_dqn_variants_forward(tmp_path):
     from scalerl_amd.trainer import OffPolicyTrainer
     for kw in (dict(dueling_dqn=True), dict(n_steps=3), dict(use_per=True),
                dict(noisy_dqn=True), dict(categorical_dqn=True),
+               dict(use_per=True, n_steps=3),  # PER+n-step compose (r2)
                dict(categorical_dqn=True, noisy_dqn=True, use_per=True)):
         args = DQNArguments(env_id="CartPole-v1", num_envs=2,
                             max_train_steps=600, warmup_learn_steps=100,
