@@ -114,3 +114,35 @@ def test_per_with_nstep_composes():
     # priorities assigned for the folded insert
     batch, idx, prio, total, pmin = p.sample_with_priorities(2)
     assert (prio > 0).all()
+
+
+def test_segment_tree_reference_api():
+    """SumSegmentTree/MinSegmentTree parity surface
+    (reference segment_tree.py:7-197): point updates, range reduce,
+    prefix-sum descent — cross-checked against numpy."""
+    import numpy as np
+    from scalerl_amd.data.segment_tree import (MinSegmentTree,
+                                               SumSegmentTree)
+    rng = np.random.default_rng(0)
+    cap = 64
+    st, mt = SumSegmentTree(cap), MinSegmentTree(cap)
+    vals = np.zeros(cap)
+    # touch every slot once (MinSegmentTree's neutral is +inf), then
+    # random overwrites
+    idxs = list(range(cap)) + [int(rng.integers(cap)) for _ in range(200)]
+    for i in idxs:
+        v = float(rng.random() * 5)
+        st[i] = v
+        mt[i] = v
+        vals[i] = v
+    assert st.sum() == pytest.approx(vals.sum())
+    assert st.sum(5, 20) == pytest.approx(vals[5:20].sum())
+    assert mt.min() == pytest.approx(vals.min())
+    assert mt.min(3, 40) == pytest.approx(vals[3:40].min())
+    assert st[7] == pytest.approx(vals[7])
+    # prefix descent: largest i with cumsum[:i] <= p
+    for p in (0.0, 0.3 * vals.sum(), 0.9999 * vals.sum()):
+        i = st.find_prefixsum_idx(p)
+        cs = np.cumsum(vals)
+        want = int(np.searchsorted(cs, p, side="right"))
+        assert i == min(want, cap - 1), (p, i, want)
