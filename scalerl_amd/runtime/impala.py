@@ -342,6 +342,12 @@ class ImpalaTrainer:
         else:
             self.shared_flat.flat.copy_(self.flat.flat)
         if self.inference == "gpu":
+            # version bumps when the D2H publish is ENQUEUED, not complete:
+            # a reader racing the copy may see a torn weight mix for one
+            # round.  That is deliberate — IMPALA only needs the BEHAVIOR
+            # logits it records to match the ones it sampled from (they
+            # always do; V-trace corrects any μ), and fencing here would
+            # put a host sync on every learn step.
             with self.weights_version.get_lock():
                 self.weights_version.value += 1
 
