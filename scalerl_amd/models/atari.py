@@ -33,9 +33,10 @@ class AtariNet(nn.Module):
         self.observation_shape = tuple(observation_shape)
         self.num_actions = num_actions
         self.use_lstm = use_lstm
-        # EXPERIMENTAL: hand-written MFMA encoder convs (ops/conv.py);
-        # requires the standard (4,84,84) shape; default off until the
-        # backward kernels are hardware-validated.
+        # Hand-written MFMA encoder convs (ops/conv.py) — ALL kernels
+        # hardware-validated (r2); opt-in because MIOpen still wins the
+        # per-op A/B (profiles/README.md; the panel/stride-decomposed v3
+        # kernels queued for r3 close the gap).  Requires (4,84,84).
         if native_conv is None:
             import os
             native_conv = bool(os.environ.get("SCALERL_NATIVE_CONV"))

@@ -1,10 +1,11 @@
-"""EXPERIMENTAL: hand-written MFMA Atari conv encoder (forward).
+"""Hand-written MFMA Atari conv encoder (fwd + wgrad + dgrad wrappers).
 
-Not yet wired into the default model paths (MIOpen via torch remains the
-production conv until these are validated on hardware — the gpu tests for
-this module are gated behind SCALERL_EXPERIMENTAL=1 so the round-end suite
-stays green).  `mfma_selftest` validates the fragment-layout constants
-first; the convs share them.
+All v2 kernels are hardware-validated (r2: 10/10 kernel-vs-oracle on
+MI355X, tests ungated).  MIOpen remains the model default until the v3
+kernels (panel-staged fwd/wgrad, stride-decomposed dgrad — written,
+CPU-math-verified, gated SCALERL_EXPERIMENTAL) win the per-op A/B
+(profiles/README.md).  `mfma_selftest` validates the fragment-layout
+constants first; every kernel shares them.
 """
 
 from __future__ import annotations
